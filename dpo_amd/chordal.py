@@ -1,0 +1,137 @@
+"""Chordal and odometry initialization.
+
+Parity: reference DPGO_utils.cpp:273-461 — B1/B2/B3 matrices of eqs.
+(69a-c) of the SE-Sync tech report; rotation chordal relaxation solved as
+a sparse least-squares problem with the first pose pinned to identity;
+per-pose SO(d) rounding; translations recovered from a second sparse
+least-squares solve. The reference uses SuiteSparse SPQR; we solve the
+(well-conditioned) normal equations with a sparse Cholesky/LU
+factorization on CPU, and a preconditioned CG path for the GPU.
+"""
+from __future__ import annotations
+
+from typing import List, Sequence, Tuple
+
+import numpy as np
+import scipy.sparse as sp
+import scipy.sparse.linalg as spla
+
+from .liegroups import project_to_rotation_group
+from .types import RelativeSEMeasurement
+
+
+def construct_b_matrices(measurements: Sequence[RelativeSEMeasurement],
+                         num_poses: int, d: int):
+    """Sparse B1 (d m x d n), B2 (d m x d^2 n), B3 (d^2 m x d^2 n)."""
+    m = len(measurements)
+    d2 = d * d
+
+    rows1: List[int] = []; cols1: List[int] = []; vals1: List[float] = []
+    rows2: List[int] = []; cols2: List[int] = []; vals2: List[float] = []
+    rows3: List[int] = []; cols3: List[int] = []; vals3: List[float] = []
+
+    for e, ms in enumerate(measurements):
+        i, j = ms.p1, ms.p2
+        sqrttau = np.sqrt(ms.tau)
+        sqrtkappa = np.sqrt(ms.kappa)
+        for l in range(d):
+            rows1 += [e * d + l, e * d + l]
+            cols1 += [i * d + l, j * d + l]
+            vals1 += [-sqrttau, sqrttau]
+        for k in range(d):
+            for rr in range(d):
+                rows2.append(d * e + rr)
+                cols2.append(d2 * i + d * k + rr)
+                vals2.append(-sqrttau * ms.t[k])
+        R = ms.R
+        for rr in range(d):
+            for c in range(d):
+                for l in range(d):
+                    rows3.append(e * d2 + d * rr + l)
+                    cols3.append(i * d2 + d * c + l)
+                    vals3.append(-sqrtkappa * R[c, rr])
+        for l in range(d2):
+            rows3.append(e * d2 + l)
+            cols3.append(j * d2 + l)
+            vals3.append(sqrtkappa)
+
+    B1 = sp.csc_matrix((vals1, (rows1, cols1)), shape=(d * m, d * num_poses))
+    B2 = sp.csc_matrix((vals2, (rows2, cols2)), shape=(d * m, d2 * num_poses))
+    B3 = sp.csc_matrix((vals3, (rows3, cols3)), shape=(d2 * m, d2 * num_poses))
+    return B1, B2, B3
+
+
+def _sparse_lsq(A: sp.spmatrix, b: np.ndarray) -> np.ndarray:
+    """min ||A x + b||_2 via normal equations with a small Tikhonov-free
+    sparse factorization (the systems here are full-rank by construction
+    once the first pose is pinned)."""
+    AtA = (A.T @ A).tocsc()
+    Atb = A.T @ b
+    try:
+        x = spla.spsolve(AtA, -Atb)
+    except RuntimeError:
+        x = spla.lsqr(A, -b, atol=1e-12, btol=1e-12, iter_lim=10000)[0]
+    return np.asarray(x)
+
+
+def chordal_initialization(d: int, num_poses: int,
+                           measurements: Sequence[RelativeSEMeasurement]
+                           ) -> np.ndarray:
+    """Returns T (d, (d+1) n): [R1 t1 R2 t2 ...] with pose 0 = identity."""
+    assert measurements, "chordal initialization needs measurements"
+    d2 = d * d
+    B1, B2, B3 = construct_b_matrices(measurements, num_poses, d)
+
+    # Rotations: pin pose 0 to I, solve for the rest.
+    B3red = B3[:, d2:]
+    Id = np.eye(d)
+    cR = B3[:, :d2] @ Id.flatten(order="F")
+    rvec = _sparse_lsq(B3red, cR)
+    Rall = np.zeros((d, d * num_poses))
+    Rall[:, :d] = Id
+    Rall[:, d:] = rvec.reshape(d, d * (num_poses - 1), order="F")
+    for i in range(1, num_poses):
+        Rall[:, i * d:(i + 1) * d] = project_to_rotation_group(
+            Rall[:, i * d:(i + 1) * d])
+
+    # Translations from the rounded rotations.
+    t = recover_translations(B1, B2, Rall)
+
+    T = np.zeros((d, num_poses * (d + 1)))
+    for i in range(num_poses):
+        T[:, i * (d + 1):i * (d + 1) + d] = Rall[:, i * d:(i + 1) * d]
+        T[:, i * (d + 1) + d] = t[:, i]
+    return T
+
+
+def recover_translations(B1: sp.spmatrix, B2: sp.spmatrix,
+                         R: np.ndarray) -> np.ndarray:
+    """Second least-squares solve for translations given rotations
+    (reference DPGO_utils.cpp:434-461)."""
+    d = R.shape[0]
+    n = R.shape[1] // d
+    rvec = R.flatten(order="F")
+    B1red = B1[:, d:]
+    c = B2 @ rvec
+    tred = _sparse_lsq(B1red, c)
+    t = np.zeros((d, n))
+    t[:, 1:] = tred.reshape(d, n - 1, order="F")
+    return t
+
+
+def odometry_initialization(d: int, num_poses: int,
+                            odometry: Sequence[RelativeSEMeasurement]
+                            ) -> np.ndarray:
+    """Dead-reckoning T_{i+1} = T_i * m_i from identity
+    (reference DPGO_utils.cpp:411-432)."""
+    dh = d + 1
+    T = np.zeros((d, num_poses * dh))
+    T[:, 0:d] = np.eye(d)
+    for k, m in enumerate(odometry):
+        src, dst = k, k + 1
+        assert m.p1 == src and m.p2 == dst, "odometry must be a chain"
+        Rsrc = T[:, src * dh:src * dh + d]
+        tsrc = T[:, src * dh + d]
+        T[:, dst * dh:dst * dh + d] = Rsrc @ m.R
+        T[:, dst * dh + d] = tsrc + Rsrc @ m.t
+    return T

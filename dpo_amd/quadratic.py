@@ -1,0 +1,395 @@
+"""Block-sparse SE(d) quadratic cost: Q (connection Laplacian) and G.
+
+The data matrix Q is assembled directly in (d+1)x(d+1) block-CSR (BSR)
+form — the layout the CDNA4 kernels consume from HBM — instead of the
+reference's scalar triplet assembly via an oriented incidence product
+A * Omega * A^T (DPGO_utils.cpp:199-271). The two agree: for an edge
+e = (i -> j) with T = [R t; 0 1] and Omega = diag(w*kappa I_d, w*tau),
+
+    Q_ii += T Omega T^T      Q_ij += -T Omega
+    Q_jj += Omega            Q_ji += -(T Omega)^T
+
+Per-agent Q additionally gets the shared-edge diagonal corrections
+(T Omega T^T at an outgoing public pose, Omega at an incoming one),
+mirroring PGOAgent::constructQMatrix (PGOAgent.cpp:720-781).
+
+The linear term G (PGOAgent::constructGMatrix, PGOAgent.cpp:783-859)
+couples local public poses to fixed neighbor poses; in the Xt layout the
+per-edge updates are Gt[p1] += -(T Omega) @ Xj_t (outgoing) and
+Gt[p2] += -(T Omega)^T @ Xi_t (incoming). Every edge contribution is
+LINEAR in the GNC weight w, so we store unit-weight blocks once and
+rebuild values by scaling — a scatter-add kernel on GPU.
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import Dict, List, Optional, Sequence, Tuple
+
+import numpy as np
+import torch
+
+from .types import RelativeSEMeasurement
+
+Tensor = torch.Tensor
+
+
+def edge_unit_blocks(m: RelativeSEMeasurement) -> Tuple[np.ndarray, np.ndarray, np.ndarray]:
+    """Unit-weight blocks (B_ii, B_jj, B_ij) of one edge's Q contribution.
+    B_ii = T Omega0 T^T, B_jj = Omega0, B_ij = -T Omega0, where Omega0
+    uses w = 1."""
+    d = m.d
+    dh = d + 1
+    TOm = np.zeros((dh, dh))
+    TOm[:d, :d] = m.kappa * m.R
+    TOm[:d, d] = m.tau * m.t
+    TOm[d, d] = m.tau
+    Bii = np.zeros((dh, dh))
+    Bii[:d, :d] = m.kappa * np.eye(d) + m.tau * np.outer(m.t, m.t)
+    Bii[:d, d] = m.tau * m.t
+    Bii[d, :d] = m.tau * m.t
+    Bii[d, d] = m.tau
+    Bjj = np.zeros((dh, dh))
+    Bjj[:d, :d] = m.kappa * np.eye(d)
+    Bjj[d, d] = m.tau
+    return Bii, Bjj, -TOm
+
+
+@dataclass
+class BSRMatrix:
+    """Symmetric block-CSR matrix with (d+1)x(d+1) fp64 tiles.
+
+    row_ptr: (n+1,) int32; col_idx: (nnzb,) int32; vals: (nnzb, dh, dh).
+    Also keeps a scalar torch CSR mirror for the CPU reference path.
+    """
+
+    n: int
+    dh: int
+    row_ptr: Tensor
+    col_idx: Tensor
+    vals: Tensor
+
+    _csr_cache: Optional[Tensor] = None
+
+    @property
+    def N(self) -> int:
+        return self.n * self.dh
+
+    def invalidate(self) -> None:
+        self._csr_cache = None
+
+    def to_scalar_csr(self) -> Tensor:
+        """Expand to a scalar torch sparse CSR tensor (CPU spmm path)."""
+        if self._csr_cache is not None:
+            return self._csr_cache
+        n, dh = self.n, self.dh
+        rp = self.row_ptr.cpu().numpy()
+        ci = self.col_idx.cpu().numpy()
+        nnzb = ci.shape[0]
+        # scalar COO indices for every element of every block
+        bro = np.repeat(np.arange(n, dtype=np.int64),
+                        np.diff(rp).astype(np.int64))  # block-row per block
+        rr = (bro[:, None, None] * dh
+              + np.arange(dh)[None, :, None]).repeat(dh, axis=2)
+        cc = (ci.astype(np.int64)[:, None, None] * dh
+              + np.arange(dh)[None, None, :]).repeat(dh, axis=1)
+        v = self.vals.cpu().numpy().reshape(nnzb, dh, dh)
+        coo = torch.sparse_coo_tensor(
+            np.stack([rr.reshape(-1), cc.reshape(-1)]),
+            v.reshape(-1), (self.N, self.N), dtype=torch.float64)
+        self._csr_cache = coo.coalesce().to_sparse_csr()
+        return self._csr_cache
+
+    def diag_slot_index(self) -> Tensor:
+        """(n,) index into vals of each diagonal block (cached)."""
+        if getattr(self, "_diag_slots", None) is None:
+            rp = self.row_ptr.cpu().numpy()
+            ci = self.col_idx.cpu().numpy()
+            import numpy as _np
+            slots = _np.zeros(self.n, dtype=_np.int64)
+            for i in range(self.n):
+                s, e = rp[i], rp[i + 1]
+                hit = _np.nonzero(ci[s:e] == i)[0]
+                slots[i] = s + (hit[0] if hit.size else 0)
+            self._diag_slots = torch.from_numpy(slots).to(self.vals.device)
+        return self._diag_slots
+
+    def diag_blocks(self) -> Tensor:
+        """(n, dh, dh) diagonal blocks."""
+        return self.vals.index_select(0, self.diag_slot_index())
+
+    def to_scipy(self):
+        """Scalar scipy CSR (host) for factorization-based preconditioners."""
+        import scipy.sparse as sp
+        csr = self.to_scalar_csr()
+        return sp.csr_matrix(
+            (csr.values().cpu().numpy(),
+             csr.col_indices().cpu().numpy(),
+             csr.crow_indices().cpu().numpy()),
+            shape=(self.N, self.N))
+
+    def to_dense(self) -> Tensor:
+        return self.to_scalar_csr().to_dense().to(self.vals.device)
+
+    def spmm(self, X: Tensor) -> Tensor:
+        """Q @ X (CPU reference via scalar CSR; the HIP path overrides)."""
+        return torch.sparse.mm(self.to_scalar_csr(), X)
+
+    def to(self, device) -> "BSRMatrix":
+        return BSRMatrix(self.n, self.dh,
+                         self.row_ptr.to(device), self.col_idx.to(device),
+                         self.vals.to(device))
+
+
+class QAssembler:
+    """Builds and incrementally re-weights a BSR connection Laplacian.
+
+    mode="full": all edges contribute all four blocks (centralized problem
+    / private measurements, constructConnectionLaplacianSE).
+    Shared edges (flagged) contribute only their local diagonal correction
+    (per-agent Q, PGOAgent.cpp:746-776).
+    """
+
+    def __init__(self, n: int, d: int,
+                 measurements: Sequence[RelativeSEMeasurement],
+                 shared_flags: Optional[Sequence[bool]] = None,
+                 local_endpoint: Optional[Sequence[int]] = None):
+        self.n, self.d = n, d
+        dh = d + 1
+        self.dh = dh
+        self.meas = list(measurements)
+        ne = len(self.meas)
+        shared = list(shared_flags) if shared_flags is not None else [False] * ne
+        lep = list(local_endpoint) if local_endpoint is not None else [0] * ne
+
+        # ---- block sparsity pattern ----------------------------------
+        pairs = set((i, i) for i in range(n))
+        for k, m in enumerate(self.meas):
+            if shared[k]:
+                continue
+            pairs.add((m.p1, m.p2))
+            pairs.add((m.p2, m.p1))
+        by_row: List[List[int]] = [[] for _ in range(n)]
+        for (i, j) in pairs:
+            by_row[i].append(j)
+        row_ptr = np.zeros(n + 1, dtype=np.int32)
+        col_idx_list: List[int] = []
+        slot: Dict[Tuple[int, int], int] = {}
+        for i in range(n):
+            for j in sorted(by_row[i]):
+                slot[(i, j)] = len(col_idx_list)
+                col_idx_list.append(j)
+            row_ptr[i + 1] = len(col_idx_list)
+        col_idx = np.array(col_idx_list, dtype=np.int32)
+        nnzb = len(col_idx_list)
+
+        # ---- per-edge unit blocks and target slots -------------------
+        # Each edge writes up to 3 distinct slots (ii, jj, ij) plus the
+        # transposed block at (j, i). We store slot ids + unit blocks flat
+        # for a single scatter-add pass (GPU-friendly).
+        slots: List[int] = []
+        blocks: List[np.ndarray] = []
+        edge_of: List[int] = []
+        transposed: List[bool] = []
+        for k, m in enumerate(self.meas):
+            Bii, Bjj, Bij = edge_unit_blocks(m)
+            if shared[k]:
+                # Diagonal correction only, at the local endpoint.
+                if lep[k] == 0:  # outgoing: local pose is p1
+                    slots.append(slot[(m.p1, m.p1)]); blocks.append(Bii)
+                else:            # incoming: local pose is p2
+                    slots.append(slot[(m.p2, m.p2)]); blocks.append(Bjj)
+                edge_of.append(k); transposed.append(False)
+            else:
+                slots.append(slot[(m.p1, m.p1)]); blocks.append(Bii)
+                edge_of.append(k); transposed.append(False)
+                slots.append(slot[(m.p2, m.p2)]); blocks.append(Bjj)
+                edge_of.append(k); transposed.append(False)
+                slots.append(slot[(m.p1, m.p2)]); blocks.append(Bij)
+                edge_of.append(k); transposed.append(False)
+                slots.append(slot[(m.p2, m.p1)]); blocks.append(Bij.T)
+                edge_of.append(k); transposed.append(True)
+
+        self._slots = torch.from_numpy(np.array(slots, dtype=np.int64))
+        self._blocks = torch.from_numpy(np.stack(blocks)) if blocks else \
+            torch.zeros(0, dh, dh, dtype=torch.float64)
+        self._edge_of = torch.from_numpy(np.array(edge_of, dtype=np.int64))
+        self._nnzb = nnzb
+        self.bsr = BSRMatrix(
+            n, dh,
+            torch.from_numpy(row_ptr), torch.from_numpy(col_idx),
+            torch.zeros(nnzb, dh, dh, dtype=torch.float64))
+
+    def assemble(self, weights: Optional[Tensor] = None) -> BSRMatrix:
+        """(Re)compute BSR values given per-edge weights (default all 1)."""
+        if weights is None:
+            weights = torch.ones(len(self.meas), dtype=torch.float64)
+        w = weights[self._edge_of]
+        vals = torch.zeros(self._nnzb, self.dh, self.dh, dtype=torch.float64)
+        vals.index_add_(0, self._slots, self._blocks * w[:, None, None])
+        self.bsr.vals = vals
+        self.bsr.invalidate()
+        return self.bsr
+
+
+def assemble_connection_laplacian(
+        measurements: Sequence[RelativeSEMeasurement], n: int, d: int,
+        weights: Optional[Sequence[float]] = None) -> BSRMatrix:
+    """Centralized / private connection Laplacian Q as BSR (parity with
+    reference constructConnectionLaplacianSE, DPGO_utils.cpp:265-271;
+    weights default to each measurement's stored weight)."""
+    qa = QAssembler(n, d, measurements)
+    if weights is None:
+        weights = [m.weight for m in measurements]
+    return qa.assemble(torch.tensor(list(weights), dtype=torch.float64))
+
+
+class GAssembler:
+    """Precomputed structure for the per-iteration linear term G.
+
+    For shared edges only. Neighbor poses arrive as a packed tensor
+    nbr (n_nbr_poses, dh, r) in a fixed slot order; assemble() performs
+    Gt[local_block] += -(w * E0) @ nbr_slot   (outgoing)
+    Gt[local_block] += -(w * E0)^T @ nbr_slot (incoming)
+    with E0 = T * Omega0. Returns dense Gt (N, r).
+    """
+
+    def __init__(self, n: int, d: int,
+                 shared_meas: Sequence[RelativeSEMeasurement],
+                 local_endpoint: Sequence[int],
+                 nbr_slot: Sequence[int]):
+        self.n, self.d = n, d
+        dh = d + 1
+        ne = len(shared_meas)
+        E0 = np.zeros((ne, dh, dh))
+        local_pose = np.zeros(ne, dtype=np.int64)
+        for k, m in enumerate(shared_meas):
+            TOm = np.zeros((dh, dh))
+            TOm[:d, :d] = m.kappa * m.R
+            TOm[:d, d] = m.tau * m.t
+            TOm[d, d] = m.tau
+            if local_endpoint[k] == 0:
+                local_pose[k] = m.p1
+                E0[k] = TOm
+            else:
+                local_pose[k] = m.p2
+                E0[k] = TOm.T
+        self.E0 = torch.from_numpy(E0)
+        self.local_pose = torch.from_numpy(local_pose)
+        self.nbr_slot = torch.tensor(list(nbr_slot), dtype=torch.int64)
+
+    def assemble(self, nbr_poses: Tensor, weights: Tensor, r: int) -> Tensor:
+        """nbr_poses: (n_slots, dh, r) packed neighbor poses (Xt blocks);
+        weights: (ne,) per-shared-edge GNC weights. Returns Gt (N, r)."""
+        dh = self.d + 1
+        dev = nbr_poses.device
+        E0 = self.E0.to(dev)
+        w = weights.to(dev)
+        Xn = nbr_poses[self.nbr_slot.to(dev)]        # (ne, dh, r)
+        contrib = -torch.bmm(E0 * w[:, None, None], Xn)
+        Gt = torch.zeros(self.n * dh, r, dtype=torch.float64, device=dev)
+        Gb = Gt.view(self.n, dh, r)
+        Gb.index_add_(0, self.local_pose.to(dev), contrib)
+        return Gt
+
+
+class QuadraticProblem:
+    """Cost oracle f(X) = 0.5 <Q, X^T X> + <X, G> on the lifted manifold.
+
+    In the Xt layout: f = 0.5 * sum((Q @ Xt) * Xt) + sum(Xt * Gt);
+    EucGrad_t = Q @ Xt + Gt; Hess-vec_t = Q @ Vt.
+    Parity: reference QuadraticProblem.cpp:50-101.
+    """
+
+    # Above this many scalar rows the dense-inverse preconditioner is
+    # replaced by block-Jacobi (memory: N^2 fp32).
+    DENSE_PRECOND_MAX_N = 24576
+
+    def __init__(self, n: int, d: int, r: int, precond: str = "auto"):
+        self.n, self.d, self.r = n, d, r
+        self.dh = d + 1
+        self.N = self.dh * n
+        self.Q: Optional[BSRMatrix] = None
+        self.Gt: Optional[Tensor] = None
+        self.precond_mode = precond
+        self._Lpre: Optional[Tensor] = None   # block-Jacobi factors
+        self._lu = None                       # CPU exact splu
+        self._Minv: Optional[Tensor] = None   # GPU dense fp32 inverse
+        from .manifold import LiftedSEManifold
+        self.manifold = LiftedSEManifold(r, d, n)
+
+    def set_q(self, Q: BSRMatrix, precond_reg: float = 0.1) -> None:
+        """Install Q and rebuild the tCG preconditioner of Q + reg I.
+
+        The reference factors Q + 0.1 I with Cholmod LDL^T once per setQ
+        (QuadraticProblem.cpp:31-42). Our modes:
+          * "exact" (CPU default): scipy splu of Q + reg I — same quality.
+          * "dense" (GPU default for N <= DENSE_PRECOND_MAX_N): explicit
+            fp32 inverse, applied as a GEMM (MFMA-friendly; avoids
+            serialized sparse triangular solves on the GPU; exactness is
+            not required of a CG preconditioner).
+          * "jacobi": (d+1)-block diagonal Cholesky (scalable fallback).
+        """
+        self.Q = Q
+        dh = self.dh
+        dev = Q.vals.device
+        mode = self.precond_mode
+        if mode == "auto":
+            if dev.type == "cpu":
+                mode = "exact"
+            else:
+                mode = "dense" if self.N <= self.DENSE_PRECOND_MAX_N else "jacobi"
+        self._active_precond = mode
+        if mode == "exact":
+            import scipy.sparse as sp
+            import scipy.sparse.linalg as spla
+            A = (Q.to_scipy() + precond_reg * sp.eye(self.N)).tocsc()
+            self._lu = spla.splu(A)
+        elif mode == "dense":
+            A = Q.to_dense()
+            A += precond_reg * torch.eye(self.N, dtype=A.dtype, device=dev)
+            L = torch.linalg.cholesky(A)
+            self._Minv = torch.cholesky_inverse(L).to(torch.float32)
+        else:
+            diag = Q.diag_blocks() + precond_reg * torch.eye(
+                dh, dtype=torch.float64, device=dev)
+            self._Lpre = torch.linalg.cholesky(diag)
+
+    def set_g(self, Gt: Tensor) -> None:
+        self.Gt = Gt
+
+    def _g(self) -> Tensor:
+        if self.Gt is None:
+            return torch.zeros(self.N, self.r, dtype=torch.float64,
+                               device=self.Q.vals.device if self.Q else "cpu")
+        return self.Gt
+
+    def f(self, X: Tensor) -> float:
+        QX = self.Q.spmm(X)
+        return float(0.5 * (QX * X).sum() + (X * self._g()).sum())
+
+    def euc_grad(self, X: Tensor) -> Tensor:
+        return self.Q.spmm(X) + self._g()
+
+    def hess_vec(self, V: Tensor) -> Tensor:
+        return self.Q.spmm(V)
+
+    def rie_grad(self, X: Tensor) -> Tensor:
+        return self.manifold.project_tangent(X, self.euc_grad(X))
+
+    def rie_grad_norm(self, X: Tensor) -> float:
+        return float(torch.linalg.norm(self.rie_grad(X)))
+
+    def precondition(self, X: Tensor, V: Tensor) -> Tensor:
+        """Apply the preconditioner then tangent-project at X
+        (reference QuadraticProblem.cpp:75-87)."""
+        mode = self._active_precond
+        if mode == "exact":
+            import torch as _t
+            Z = _t.from_numpy(self._lu.solve(V.cpu().numpy()))
+            Z = Z.to(V.device)
+        elif mode == "dense":
+            Z = (self._Minv @ V.to(torch.float32)).to(torch.float64)
+        else:
+            Vb = V.view(self.n, self.dh, self.r)
+            Z = torch.cholesky_solve(Vb, self._Lpre).reshape(self.N, self.r)
+        return self.manifold.project_tangent(X, Z)
