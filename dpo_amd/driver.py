@@ -56,6 +56,8 @@ class MultiRobotDriver:
                  partition: str | Sequence[int] = "contiguous",
                  acceleration: bool = False,
                  robust: RobustCostType = RobustCostType.L2,
+                 robust_params=None,
+                 robust_inner_iters: int = 30,
                  device: str = "cpu",
                  verbose: bool = False,
                  selection: str = "greedy"):
@@ -96,7 +98,10 @@ class MultiRobotDriver:
             p = PGOAgentParams(d=d, r=r, num_robots=num_robots,
                                acceleration=acceleration,
                                robust_cost_type=robust,
+                               robust_opt_inner_iters=robust_inner_iters,
                                verbose=verbose, device=device)
+            if robust_params is not None:
+                p.robust_cost_params = robust_params
             a = PGOAgent(rb, p)
             if rb > 0:
                 a.set_lifting_matrix(self.agents[0].get_lifting_matrix())
@@ -114,6 +119,21 @@ class MultiRobotDriver:
                 Xr[:, i * self.dh:(i + 1) * self.dh] = \
                     X_chordal[:, g * self.dh:(g + 1) * self.dh]
             self.agents[rb].set_x(Xr)
+
+        # Agent-quotient-graph coloring for the "colored" selection rule:
+        # agents of one color share no edge, so their simultaneous block
+        # updates compose an exact block Gauss-Seidel sweep (the
+        # scalable multi-GPU schedule).
+        colors = [-1] * num_robots
+        for rb in range(num_robots):
+            used = {colors[nb] for nb in self.agents[rb].get_neighbors()
+                    if colors[nb] >= 0}
+            c = 0
+            while c in used:
+                c += 1
+            colors[rb] = c
+        self._colors = colors
+        self._num_colors = max(colors) + 1 if colors else 1
 
         self._Xopt = torch.zeros(self.dh * num_poses, r,
                                  dtype=torch.float64,
@@ -165,10 +185,25 @@ class MultiRobotDriver:
             for it in range(max_iters):
                 if self.selection == "parallel":
                     # dpo_amd extension: all agents exchange + optimize
-                    # concurrently each round (uniform block updates).
+                    # concurrently each round (Jacobi-style; may stall on
+                    # tightly coupled partitions — prefer "colored").
                     for a in self.agents:
                         self._exchange_with(a.id, acceleration)
                     for a in self.agents:
+                        a.iterate(True)
+                elif self.selection == "colored":
+                    # dpo_amd extension: agents of the active color (a
+                    # shared-edge independent set) update concurrently =
+                    # exact block Gauss-Seidel; scales to one agent/GPU.
+                    color = it % self._num_colors
+                    active = [a for a in self.agents
+                              if self._colors[a.id] == color]
+                    for a in self.agents:
+                        if self._colors[a.id] != color:
+                            a.iterate(False)
+                    for a in active:
+                        self._exchange_with(a.id, acceleration)
+                    for a in active:
                         a.iterate(True)
                 else:
                     for a in self.agents:
