@@ -512,11 +512,30 @@ class PGOAgent:
                 device=self.device))
 
         # RBCD knob set (PGOAgent.cpp:1134-1137).
+        X_start = self.Y if acceleration else self.X
+        if (self.device.type != "cpu"
+                and self.params.algorithm == OptAlgorithm.RTR):
+            # Device-resident solve: tCG control state lives on the GPU,
+            # one host sync per local solve.
+            if getattr(self, "_dev_solver", None) is None:
+                from .ops.hip_backend import DeviceSolver
+                self._dev_solver = DeviceSolver(self.n, self.d, self.r,
+                                                self.device, max_inner=10)
+            X_work = X_start.clone() if acceleration else self.X
+            stats = self._dev_solver.solve(self.problem, X_work,
+                                           tol=1e-2, Delta0=100.0)
+            self.X = X_work
+            from .types import OptResult
+            res = OptResult(success=True,
+                            f_init=stats["f_init"], f_opt=stats["f_opt"],
+                            grad_norm_init=stats["grad_norm_init"],
+                            grad_norm_opt=stats.get("grad_norm_opt", 0.0))
+            self.last_opt_result = res
+            return True
         tr = TRParams(tolerance=1e-2, initial_radius=100.0,
                       max_iterations=1, max_inner_iterations=10)
         opt = QuadraticOptimizer(self.problem, self.params.algorithm, tr,
                                  verbose=self.params.verbose)
-        X_start = self.Y if acceleration else self.X
         self.X = opt.optimize(X_start)
         self.last_opt_result = opt.result
         return True

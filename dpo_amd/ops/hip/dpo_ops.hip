@@ -1,0 +1,925 @@
+// dpo_amd HIP/CDNA4 kernels (gfx950 / MI355X) — fp64 core math.
+//
+// Design notes (MI355X-first):
+//  * The per-agent pose-graph problems are small (N = (d+1)n scalar rows,
+//    typically 1e3..1e5; the whole working set is L2/LLC-resident), so the
+//    kernels are optimized for LATENCY and FUSION, not TFLOPs: every
+//    elementwise/projection pass fuses its reductions (atomicAdd into a
+//    device control block) so the truncated-CG loop needs no host round
+//    trips (the C++ orchestrator enqueues guarded kernels; control
+//    decisions run in single-wave kernels on-device).
+//  * Layout: X is (N, r) row-major fp64 ("Xt layout"): pose i occupies
+//    rows [i*dh, i*dh+dh), the first d of which are the transposed
+//    Stiefel block, the last the translation. Consecutive lanes read
+//    consecutive (row, col) elements -> coalesced.
+//  * Q is (d+1)x(d+1) block-CSR ("BSR"): row_ptr (n+1), col_idx (nnzb),
+//    vals (nnzb, dh, dh). The SpMM assigns a pose-block row per
+//    dh*r-thread group; fp64 FMA; bandwidth/latency-bound at these sizes.
+//  * Wavefront = 64; blocks are multiples of 64 threads.
+//
+// Functional parity targets: Hessian-vec V*Q
+// (reference QuadraticProblem.cpp:68-73), tangent projection / polar
+// retraction (ROPTLIB Stiefel ops), preconditioner apply
+// (QuadraticProblem.cpp:75-87), G/Q assembly (PGOAgent.cpp:720-859).
+
+#include <hip/hip_runtime.h>
+#include <math.h>
+#include <stdint.h>
+#include <stdio.h>
+
+#define DPO_CHECK(x)                                                     \
+  do {                                                                   \
+    hipError_t _e = (x);                                                 \
+    if (_e != hipSuccess) {                                              \
+      fprintf(stderr, "HIP error %s at %s:%d\n", hipGetErrorString(_e),  \
+              __FILE__, __LINE__);                                       \
+    }                                                                    \
+  } while (0)
+
+// ---------------------------------------------------------------------
+// control block layout (fp64 slots) shared with the C++ orchestrator
+// ---------------------------------------------------------------------
+enum CtrlSlot {
+  C_STATUS = 0,   // 0 run, 1 tcg_stop, 2 accepted, 3 give_up, 4 no_update
+  C_FX = 1,
+  C_GN0SQ = 2,
+  C_RR = 3,
+  C_ZR = 4,
+  C_EPE = 5,
+  C_EPD = 6,
+  C_DPD = 7,
+  C_COEF = 8,     // alpha or tau applied in the current eta update
+  C_BETA = 9,
+  C_ITER = 10,
+  C_J = 11,       // snapshots stored (completed tCG iterations)
+  C_RADIUS = 12,
+  C_FPROP = 13,
+  C_DM = 14,      // model decrease of current candidate step
+  C_JSTAR = 15,
+  C_TAUSTAR = 16,
+  C_USE_CURRENT = 17,  // candidate step == current eta (converged tCG)
+  C_STOP_PENDING = 18, // boundary/negcurv hit in current iteration
+  C_BOUND = 19,        // tCG residual bound
+  C_GN1SQ = 20,
+  C_RHO = 21,
+  C_HLEN = 22,     // valid Krylov-history entries (radius-independent)
+  C_DOT0 = 24,   // scratch dot slots (cleared by ctrl kernels)
+  C_DOT1 = 25,
+  C_DOT2 = 26,
+  C_DOT3 = 27,
+  C_HIST = 32,   // 5 arrays of length MAX_TCG: z_r, d_Hd, e_Pe, e_Pd, d_Pd
+};
+#define MAX_TCG 16
+#define H_ZR(j) (C_HIST + (j))
+#define H_DHD(j) (C_HIST + MAX_TCG + (j))
+#define H_EPE(j) (C_HIST + 2 * MAX_TCG + (j))
+#define H_EPD(j) (C_HIST + 3 * MAX_TCG + (j))
+#define H_DPD(j) (C_HIST + 4 * MAX_TCG + (j))
+#define CTRL_SIZE (C_HIST + 5 * MAX_TCG)
+
+enum Status {
+  ST_RUN = 0,
+  ST_TCG_STOP = 1,
+  ST_ACCEPTED = 2,
+  ST_GIVE_UP = 3,
+  ST_NO_UPDATE = 4,
+};
+
+// guard < 0: no guard; else kernel runs only when status == guard.
+__device__ __forceinline__ bool guarded_off(const double* ctrl, int guard) {
+  return guard >= 0 && ctrl != nullptr && ctrl[C_STATUS] != (double)guard;
+}
+
+// wave-level + block-level reduction, then one atomicAdd per block
+__device__ __forceinline__ void block_reduce_atomic(double v, double* dst) {
+  __shared__ double sh[16];  // up to 1024/64 waves
+  int lane = threadIdx.x & 63;
+  int wave = threadIdx.x >> 6;
+  for (int off = 32; off > 0; off >>= 1)
+    v += __shfl_down(v, off, 64);
+  if (lane == 0) sh[wave] = v;
+  __syncthreads();
+  int nwaves = (blockDim.x + 63) >> 6;
+  if (wave == 0) {
+    double s = (lane < nwaves) ? sh[lane] : 0.0;
+    for (int off = 32; off > 0; off >>= 1)
+      s += __shfl_down(s, off, 64);
+    if (lane == 0 && s != 0.0) atomicAdd(dst, s);
+  }
+}
+
+// ---------------------------------------------------------------------
+// BSR SpMM: out(N, r) = Q(N, N) @ X(N, r)
+// One dh*r-thread group per pose-block row; each thread owns one
+// (row_in_block, col) output element and accumulates over the row's
+// blocks. Q block values are broadcast from L2 (tiny per-agent Q is
+// cache-resident); X block reads are contiguous dh*r segments.
+// ---------------------------------------------------------------------
+__global__ void k_bsr_spmm(const int* __restrict__ row_ptr,
+                           const int* __restrict__ col_idx,
+                           const double* __restrict__ vals,
+                           const double* __restrict__ X,
+                           double* __restrict__ out,
+                           int n, int dh, int r,
+                           const double* __restrict__ ctrl, int guard) {
+  if (guarded_off(ctrl, guard)) return;
+  const int tile = dh * r;
+  const int per_block = blockDim.x / tile;
+  const int slot = threadIdx.x / tile;
+  const int t = threadIdx.x % tile;
+  const int i = blockIdx.x * per_block + slot;
+  if (threadIdx.x >= per_block * tile || i >= n) return;
+  const int c = t / r;   // row inside block
+  const int k = t % r;   // column of X
+  const int s = row_ptr[i], e = row_ptr[i + 1];
+  double acc = 0.0;
+  for (int p = s; p < e; ++p) {
+    const int j = col_idx[p];
+    const double* B = vals + (size_t)p * dh * dh;
+    const double* Xj = X + (size_t)j * dh * r;
+    #pragma unroll 4
+    for (int cc = 0; cc < dh; ++cc)
+      acc = fma(B[c * dh + cc], Xj[cc * r + k], acc);
+  }
+  out[(size_t)i * dh * r + c * r + k] = acc;
+}
+
+// ---------------------------------------------------------------------
+// Fused tangent projection + dots.
+//   P = V (+ G) projected onto T_X; optionally negate; write to out.
+//   Optional atomic dots: <P, dotWith> -> ctrl[dot_slot]
+//                         <V+G, X> -> ctrl[dot_slot2]   (for f(X))
+// One thread per pose; all of Yt (d x r), Vt (dh x r) live in registers
+// (r <= DPO_MAX_R, dh <= 4).
+// ---------------------------------------------------------------------
+#define DPO_MAX_R 8
+#define DPO_MAX_DH 4
+
+template <int NEG>
+__global__ void k_proj_dots(const double* __restrict__ X,
+                            const double* __restrict__ V,
+                            const double* __restrict__ G,
+                            double* __restrict__ out,
+                            const double* __restrict__ dotWith,
+                            double* __restrict__ ctrl,
+                            int n, int d, int r,
+                            int dot_slot, int dot_slot2,
+                            int guard) {
+  if (guarded_off(ctrl, guard)) return;
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  const int dh = d + 1;
+  double Yt[DPO_MAX_DH - 1][DPO_MAX_R];
+  double Vt[DPO_MAX_DH][DPO_MAX_R];
+  double dot_vx = 0.0;
+  if (i < n) {
+    const double* Xi = X + (size_t)i * dh * r;
+    const double* Vi = V + (size_t)i * dh * r;
+    const double* Gi = G ? G + (size_t)i * dh * r : nullptr;
+    for (int c = 0; c < dh; ++c)
+      for (int k = 0; k < r; ++k) {
+        double v = Vi[c * r + k];
+        if (Gi) v += Gi[c * r + k];
+        Vt[c][k] = v;
+        if (dot_slot2 >= 0) dot_vx = fma(v, Xi[c * r + k], dot_vx);
+      }
+    for (int c = 0; c < d; ++c)
+      for (int k = 0; k < r; ++k)
+        Yt[c][k] = Xi[c * r + k];
+    // S = sym(Yt Vt^T)  (d x d)
+    double S[DPO_MAX_DH - 1][DPO_MAX_DH - 1];
+    for (int a = 0; a < d; ++a)
+      for (int b = 0; b < d; ++b) {
+        double s = 0.0;
+        for (int k = 0; k < r; ++k) s = fma(Yt[a][k], Vt[b][k], s);
+        S[a][b] = s;
+      }
+    for (int a = 0; a < d; ++a)
+      for (int b = a; b < d; ++b) {
+        double s = 0.5 * (S[a][b] + S[b][a]);
+        S[a][b] = s;
+        S[b][a] = s;
+      }
+    // P = Vt - S Yt on Stiefel rows
+    for (int a = 0; a < d; ++a)
+      for (int k = 0; k < r; ++k) {
+        double acc = Vt[a][k];
+        for (int b = 0; b < d; ++b) acc = fma(-S[a][b], Yt[b][k], acc);
+        Vt[a][k] = acc;
+      }
+  }
+  double dot_pw = 0.0;
+  if (i < n) {
+    double* Oi = out + (size_t)i * dh * r;
+    const double* Wi = dotWith ? dotWith + (size_t)i * dh * r : nullptr;
+    for (int c = 0; c < dh; ++c)
+      for (int k = 0; k < r; ++k) {
+        double p = NEG ? -Vt[c][k] : Vt[c][k];
+        Oi[c * r + k] = p;
+        if (dot_slot >= 0) {
+          double w = Wi ? Wi[c * r + k] : Vt[c][k];  // default: <P,P>
+          dot_pw = fma(p, w, dot_pw);
+        }
+      }
+  }
+  if (dot_slot >= 0) block_reduce_atomic(dot_pw, ctrl + dot_slot);
+  if (dot_slot2 >= 0) block_reduce_atomic(dot_vx, ctrl + dot_slot2);
+}
+
+// ---------------------------------------------------------------------
+// Batched polar projection onto (St(d, r) x R^r)^n of an affine
+// combination  M = ca*A + cb*B + cc*C  (B, C optional).
+// polar(Mt) for the wide d x r Stiefel block via the analytic
+// eigendecomposition of the d x d Gram matrix (d <= 3).
+// Translations pass through the affine combination unchanged.
+// One thread per pose, everything in registers.
+// ---------------------------------------------------------------------
+__device__ void sym_eig(const double S[3][3], int d, double lam[3],
+                        double Q[3][3]) {
+  if (d == 1) {
+    lam[0] = S[0][0];
+    Q[0][0] = 1.0;
+    return;
+  }
+  if (d == 2) {
+    double tr = S[0][0] + S[1][1];
+    double det = S[0][0] * S[1][1] - S[0][1] * S[1][0];
+    double disc = sqrt(fmax(tr * tr * 0.25 - det, 0.0));
+    lam[0] = tr * 0.5 + disc;
+    lam[1] = tr * 0.5 - disc;
+    // eigenvector for lam[0]
+    double a = S[0][0] - lam[1], b = S[0][1];
+    double n0 = hypot(a, S[1][0]);
+    if (n0 > 1e-300) {
+      Q[0][0] = a / n0; Q[1][0] = S[1][0] / n0;
+    } else {
+      Q[0][0] = 1.0; Q[1][0] = 0.0;
+    }
+    Q[0][1] = -Q[1][0];
+    Q[1][1] = Q[0][0];
+    (void)b;
+    return;
+  }
+  // d == 3: trigonometric (Smith) method + one inverse-iteration polish.
+  double p1 = S[0][1] * S[0][1] + S[0][2] * S[0][2] + S[1][2] * S[1][2];
+  double q = (S[0][0] + S[1][1] + S[2][2]) / 3.0;
+  double a00 = S[0][0] - q, a11 = S[1][1] - q, a22 = S[2][2] - q;
+  double p2 = a00 * a00 + a11 * a11 + a22 * a22 + 2.0 * p1;
+  double p = sqrt(fmax(p2 / 6.0, 0.0));
+  if (p < 1e-300) {
+    lam[0] = lam[1] = lam[2] = q;
+    for (int i = 0; i < 3; ++i)
+      for (int j = 0; j < 3; ++j) Q[i][j] = (i == j) ? 1.0 : 0.0;
+    return;
+  }
+  double invp = 1.0 / p;
+  // B = (S - q I) / p ; r = det(B)/2
+  double B[3][3];
+  B[0][0] = a00 * invp; B[0][1] = S[0][1] * invp; B[0][2] = S[0][2] * invp;
+  B[1][0] = B[0][1];    B[1][1] = a11 * invp;     B[1][2] = S[1][2] * invp;
+  B[2][0] = B[0][2];    B[2][1] = B[1][2];        B[2][2] = a22 * invp;
+  double detB = B[0][0] * (B[1][1] * B[2][2] - B[1][2] * B[2][1])
+              - B[0][1] * (B[1][0] * B[2][2] - B[1][2] * B[2][0])
+              + B[0][2] * (B[1][0] * B[2][1] - B[1][1] * B[2][0]);
+  double rr = fmin(fmax(detB * 0.5, -1.0), 1.0);
+  double phi = acos(rr) / 3.0;
+  lam[0] = q + 2.0 * p * cos(phi);
+  lam[2] = q + 2.0 * p * cos(phi + 2.0943951023931953);  // + 2pi/3
+  lam[1] = 3.0 * q - lam[0] - lam[2];
+  // eigenvectors: v_k = (S - lam_a I)(S - lam_b I) e_col, pick best col
+  for (int k = 0; k < 3; ++k) {
+    double la = lam[(k + 1) % 3], lb = lam[(k + 2) % 3];
+    double M[3][3];
+    for (int i = 0; i < 3; ++i)
+      for (int j = 0; j < 3; ++j) {
+        double acc = 0.0;
+        for (int m = 0; m < 3; ++m) {
+          double s_im = S[i][m] - (i == m ? la : 0.0);
+          double s_mj = S[m][j] - (m == j ? lb : 0.0);
+          acc += s_im * s_mj;
+        }
+        M[i][j] = acc;
+      }
+    // column with largest norm
+    int bc = 0; double bn = -1.0;
+    for (int j = 0; j < 3; ++j) {
+      double nn = M[0][j] * M[0][j] + M[1][j] * M[1][j] + M[2][j] * M[2][j];
+      if (nn > bn) { bn = nn; bc = j; }
+    }
+    double nv = sqrt(fmax(bn, 1e-300));
+    Q[0][k] = M[0][bc] / nv;
+    Q[1][k] = M[1][bc] / nv;
+    Q[2][k] = M[2][bc] / nv;
+  }
+  // re-orthogonalize (degenerate eigenvalues): Gram-Schmidt
+  for (int k = 1; k < 3; ++k)
+    for (int m = 0; m < k; ++m) {
+      double dp = Q[0][k] * Q[0][m] + Q[1][k] * Q[1][m] + Q[2][k] * Q[2][m];
+      Q[0][k] -= dp * Q[0][m];
+      Q[1][k] -= dp * Q[1][m];
+      Q[2][k] -= dp * Q[2][m];
+      double nn = sqrt(Q[0][k] * Q[0][k] + Q[1][k] * Q[1][k]
+                       + Q[2][k] * Q[2][k]);
+      if (nn > 1e-300) {
+        Q[0][k] /= nn; Q[1][k] /= nn; Q[2][k] /= nn;
+      }
+    }
+}
+
+__global__ void k_polar_affine(const double* __restrict__ A,
+                               const double* __restrict__ B,
+                               const double* __restrict__ C,
+                               double ca, double cb, double cc,
+                               double* __restrict__ out,
+                               int n, int d, int r,
+                               const double* __restrict__ ctrl, int guard) {
+  if (guarded_off(ctrl, guard)) return;
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  const int dh = d + 1;
+  double Mt[DPO_MAX_DH][DPO_MAX_R];
+  const double* Ai = A + (size_t)i * dh * r;
+  const double* Bi = B ? B + (size_t)i * dh * r : nullptr;
+  const double* Ci = C ? C + (size_t)i * dh * r : nullptr;
+  for (int c = 0; c < dh; ++c)
+    for (int k = 0; k < r; ++k) {
+      double v = ca * Ai[c * r + k];
+      if (Bi) v = fma(cb, Bi[c * r + k], v);
+      if (Ci) v = fma(cc, Ci[c * r + k], v);
+      Mt[c][k] = v;
+    }
+  // Gram = Mt Mt^T (d x d); polar(Mt) = Gram^{-1/2} Mt
+  double S[3][3] = {};
+  for (int a = 0; a < d; ++a)
+    for (int b = 0; b < d; ++b) {
+      double s = 0.0;
+      for (int k = 0; k < r; ++k) s = fma(Mt[a][k], Mt[b][k], s);
+      S[a][b] = s;
+    }
+  double lam[3], Q[3][3];
+  sym_eig(S, d, lam, Q);
+  // G^{-1/2} = Q diag(lam^{-1/2}) Q^T with clamping
+  double lmax = fmax(lam[0], 1e-300);
+  double Gi[3][3];
+  for (int a = 0; a < d; ++a)
+    for (int b = 0; b < d; ++b) {
+      double s = 0.0;
+      for (int k = 0; k < d; ++k) {
+        double l = fmax(lam[k], 1e-14 * lmax);
+        s += Q[a][k] * Q[b][k] * rsqrt(l);
+      }
+      Gi[a][b] = s;
+    }
+  double* Oi = out + (size_t)i * dh * r;
+  for (int a = 0; a < d; ++a)
+    for (int k = 0; k < r; ++k) {
+      double s = 0.0;
+      for (int b = 0; b < d; ++b) s = fma(Gi[a][b], Mt[b][k], s);
+      Oi[a * r + k] = s;
+    }
+  for (int k = 0; k < r; ++k) Oi[d * r + k] = Mt[d][k];
+}
+
+// ---------------------------------------------------------------------
+// Dense fp32 preconditioner apply: Z(N, r) = Minv(N, N) @ V(N, r).
+// Minv is symmetric -> read column-major (Minv[j*N+i]) so consecutive
+// threads (i) are coalesced; j-loop broadcasts V[j][*] via LDS.
+// fp32 storage halves the bandwidth; accumulate fp64.
+// ---------------------------------------------------------------------
+__global__ void k_precond_dense(const float* __restrict__ Minv,
+                                const double* __restrict__ V,
+                                double* __restrict__ Z,
+                                int N, int r,
+                                const double* __restrict__ ctrl, int guard) {
+  if (guarded_off(ctrl, guard)) return;
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  extern __shared__ double shv[];  // TILE_J * r
+  const int TILE_J = 64;
+  double acc[DPO_MAX_R] = {};
+  for (int j0 = 0; j0 < N; j0 += TILE_J) {
+    const int jn = min(TILE_J, N - j0);
+    __syncthreads();
+    for (int t = threadIdx.x; t < jn * r; t += blockDim.x)
+      shv[t] = V[(size_t)(j0 + t / r) * r + (t % r)];
+    __syncthreads();
+    if (i < N) {
+      for (int j = 0; j < jn; ++j) {
+        const double m = (double)Minv[(size_t)(j0 + j) * N + i];
+        #pragma unroll
+        for (int k = 0; k < DPO_MAX_R; ++k)
+          if (k < r) acc[k] = fma(m, shv[j * r + k], acc[k]);
+      }
+    }
+  }
+  if (i < N)
+    for (int k = 0; k < r; ++k) Z[(size_t)i * r + k] = acc[k];
+}
+
+// Block-Jacobi apply: per pose solve (L L^T) z = v with stored Cholesky
+// factors L (n, dh, dh). One thread per pose per rhs column.
+__global__ void k_precond_jacobi(const double* __restrict__ L,
+                                 const double* __restrict__ V,
+                                 double* __restrict__ Z,
+                                 int n, int dh, int r,
+                                 const double* __restrict__ ctrl, int guard) {
+  if (guarded_off(ctrl, guard)) return;
+  const int idx = blockIdx.x * blockDim.x + threadIdx.x;
+  if (idx >= n * r) return;
+  const int i = idx / r, k = idx % r;
+  const double* Li = L + (size_t)i * dh * dh;
+  double y[DPO_MAX_DH];
+  // forward solve L y = v
+  for (int a = 0; a < dh; ++a) {
+    double s = V[((size_t)i * dh + a) * r + k];
+    for (int b = 0; b < a; ++b) s = fma(-Li[a * dh + b], y[b], s);
+    y[a] = s / Li[a * dh + a];
+  }
+  // backward solve L^T z = y
+  for (int a = dh - 1; a >= 0; --a) {
+    double s = y[a];
+    for (int b = a + 1; b < dh; ++b) s = fma(-Li[b * dh + a], y[b], s);
+    y[a] = s / Li[a * dh + a];
+  }
+  for (int a = 0; a < dh; ++a) Z[((size_t)i * dh + a) * r + k] = y[a];
+}
+
+// ---------------------------------------------------------------------
+// Fused tCG vector updates (guarded; coefficients read from ctrl):
+//   snapshot: eta_snap[j] = eta, delta_snap[j] = delta  (pre-update)
+//   eta += coef * delta;  if (!stop_pending) { r += alpha*Hd; rr+=r^2 }
+// ---------------------------------------------------------------------
+__global__ void k_tcg_update(double* __restrict__ eta,
+                             double* __restrict__ rvec,
+                             const double* __restrict__ delta,
+                             const double* __restrict__ Hd,
+                             double* __restrict__ eta_snap,
+                             double* __restrict__ delta_snap,
+                             double* __restrict__ ctrl,
+                             long total) {
+  if (guarded_off(ctrl, ST_RUN)) return;
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const double coef = ctrl[C_COEF];
+  const int stop_pending = (int)ctrl[C_STOP_PENDING];
+  const int j = (int)ctrl[C_ITER];
+  double rr = 0.0;
+  if (i < total) {
+    const double dl = delta[i];
+    const double et = eta[i];
+    eta_snap[(size_t)j * total + i] = et;
+    delta_snap[(size_t)j * total + i] = dl;
+    eta[i] = fma(coef, dl, et);
+    if (!stop_pending) {
+      const double rn = fma(coef, Hd[i], rvec[i]);
+      rvec[i] = rn;
+      rr = rn * rn;
+    }
+  }
+  if (!stop_pending) block_reduce_atomic(rr, ctrl + C_DOT1);
+}
+
+// delta = -z + beta * delta
+__global__ void k_tcg_delta(double* __restrict__ delta,
+                            const double* __restrict__ z,
+                            double* __restrict__ ctrl, long total) {
+  if (guarded_off(ctrl, ST_RUN)) return;
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= total) return;
+  delta[i] = fma(ctrl[C_BETA], delta[i], -z[i]);
+}
+
+// step = eta_snap[jstar] + tau*delta_snap[jstar], or current eta
+__global__ void k_form_step(double* __restrict__ step,
+                            const double* __restrict__ eta,
+                            const double* __restrict__ eta_snap,
+                            const double* __restrict__ delta_snap,
+                            const double* __restrict__ ctrl, long total) {
+  if (guarded_off(ctrl, ST_TCG_STOP)) return;
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= total) return;
+  if (ctrl[C_USE_CURRENT] != 0.0) {
+    step[i] = eta[i];
+  } else {
+    const int j = (int)ctrl[C_JSTAR];
+    const double tau = ctrl[C_TAUSTAR];
+    step[i] = fma(tau, delta_snap[(size_t)j * total + i],
+                  eta_snap[(size_t)j * total + i]);
+  }
+}
+
+// out = a*A + b*B elementwise (unguarded helper)
+__global__ void k_axpby(const double* __restrict__ A,
+                        const double* __restrict__ B,
+                        double a, double b,
+                        double* __restrict__ out, long total) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= total) return;
+  double v = a * A[i];
+  if (B) v = fma(b, B[i], v);
+  out[i] = v;
+}
+
+// dots: ctrl[slot] += <A, B>, optionally ctrl[slot2] += <A, C>
+__global__ void k_dots(const double* __restrict__ A,
+                       const double* __restrict__ B,
+                       const double* __restrict__ C,
+                       double* __restrict__ ctrl,
+                       int slot, int slot2, long total, int guard) {
+  if (guarded_off(ctrl, guard)) return;
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  double d0 = 0.0, d1 = 0.0;
+  if (i < total) {
+    const double a = A[i];
+    d0 = a * B[i];
+    if (C) d1 = a * C[i];
+  }
+  block_reduce_atomic(d0, ctrl + slot);
+  if (C && slot2 >= 0) block_reduce_atomic(d1, ctrl + slot2);
+}
+
+// ---------------------------------------------------------------------
+// Single-wave tCG control kernels (device-side branch logic)
+// ---------------------------------------------------------------------
+__global__ void k_ctrl_init(double* ctrl, double tol, double Delta0,
+                            double theta, double kappa) {
+  if (threadIdx.x != 0) return;
+  // C_DOT0 = <QX + G, X>, C_DOT2 = <G, X>
+  // => f = 0.5 <QX, X> + <G, X> = 0.5 (C_DOT0 + C_DOT2)
+  const double fX = 0.5 * (ctrl[C_DOT0] + ctrl[C_DOT2]);
+  const double gn0sq = ctrl[C_DOT1];
+  ctrl[C_FX] = fX;
+  ctrl[C_GN0SQ] = gn0sq;
+  ctrl[C_RR] = gn0sq;
+  const double norm_r0 = sqrt(gn0sq);
+  ctrl[C_BOUND] = norm_r0 * fmin(pow(norm_r0, theta), kappa);
+  ctrl[C_RADIUS] = Delta0;
+  ctrl[C_EPE] = 0.0;
+  ctrl[C_EPD] = 0.0;
+  ctrl[C_ITER] = 0.0;
+  ctrl[C_J] = 0.0;
+  ctrl[C_USE_CURRENT] = 0.0;
+  ctrl[C_STOP_PENDING] = 0.0;
+  ctrl[C_BETA] = 0.0;
+  ctrl[C_STATUS] = (norm_r0 < tol) ? (double)ST_NO_UPDATE : (double)ST_RUN;
+  ctrl[C_DOT0] = 0.0;
+  ctrl[C_DOT1] = 0.0;
+  ctrl[C_DOT2] = 0.0;
+  ctrl[C_DOT3] = 0.0;
+}
+
+// after z0 projection+dot (C_DOT0 = <z0, r0>)
+__global__ void k_ctrl_z0(double* ctrl) {
+  if (threadIdx.x != 0) return;
+  if (ctrl[C_STATUS] != (double)ST_RUN) return;
+  const double zr = ctrl[C_DOT0];
+  ctrl[C_ZR] = zr;
+  ctrl[C_DPD] = zr;
+  ctrl[C_DOT0] = 0.0;
+}
+
+// after Hd projection+dot (C_DOT0 = <delta, Hd>): alpha / boundary logic
+__global__ void k_ctrl_alpha(double* ctrl) {
+  if (threadIdx.x != 0) return;
+  if (ctrl[C_STATUS] != (double)ST_RUN) return;
+  const int j = (int)ctrl[C_ITER];
+  const double d_Hd = ctrl[C_DOT0];
+  ctrl[C_DOT0] = 0.0;
+  const double z_r = ctrl[C_ZR];
+  const double e_Pe = ctrl[C_EPE], e_Pd = ctrl[C_EPD], d_Pd = ctrl[C_DPD];
+  const double radius = ctrl[C_RADIUS];
+  // scalar history for radius-replay (shrink loop reuses the Krylov path)
+  ctrl[H_ZR(j)] = z_r;
+  ctrl[H_DHD(j)] = d_Hd;
+  ctrl[H_EPE(j)] = e_Pe;
+  ctrl[H_EPD(j)] = e_Pd;
+  ctrl[H_DPD(j)] = d_Pd;
+  const double alpha = z_r / d_Hd;
+  const double e_Pe_new = e_Pe + 2.0 * alpha * e_Pd + alpha * alpha * d_Pd;
+  if (d_Hd <= 0.0 || e_Pe_new >= radius * radius) {
+    const double disc = e_Pd * e_Pd + d_Pd * (radius * radius - e_Pe);
+    const double tau = (d_Pd > 0.0)
+        ? (-e_Pd + sqrt(fmax(disc, 0.0))) / d_Pd : 0.0;
+    ctrl[C_COEF] = tau;
+    ctrl[C_STOP_PENDING] = 1.0;
+  } else {
+    ctrl[C_COEF] = alpha;
+    ctrl[C_EPE] = e_Pe_new;
+    ctrl[C_STOP_PENDING] = 0.0;
+  }
+}
+
+// after k_tcg_update (C_DOT1 = ||r||^2 when continuing)
+__global__ void k_ctrl_rr(double* ctrl) {
+  if (threadIdx.x != 0) return;
+  if (ctrl[C_STATUS] != (double)ST_RUN) return;
+  const int j = (int)ctrl[C_ITER];
+  if (ctrl[C_STOP_PENDING] != 0.0) {
+    ctrl[C_STATUS] = (double)ST_TCG_STOP;
+    ctrl[C_J] = (double)j;         // truncated at snapshot j
+    ctrl[C_HLEN] = (double)(j + 1);
+    ctrl[C_USE_CURRENT] = 0.0;
+    return;
+  }
+  const double rr = ctrl[C_DOT1];
+  ctrl[C_DOT1] = 0.0;
+  ctrl[C_RR] = rr;
+  if (sqrt(rr) <= ctrl[C_BOUND]) {
+    ctrl[C_STATUS] = (double)ST_TCG_STOP;
+    ctrl[C_J] = (double)(j + 1);
+    ctrl[C_HLEN] = (double)(j + 1);
+    ctrl[C_USE_CURRENT] = 1.0;     // converged: step = current eta
+  }
+}
+
+// after z projection+dot (C_DOT0 = <z, r>): beta and delta scalars
+__global__ void k_ctrl_beta(double* ctrl) {
+  if (threadIdx.x != 0) return;
+  if (ctrl[C_STATUS] != (double)ST_RUN) return;
+  const double z_r_new = ctrl[C_DOT0];
+  ctrl[C_DOT0] = 0.0;
+  const double z_r = ctrl[C_ZR];
+  const double alpha = ctrl[C_COEF];
+  const double beta = z_r_new / z_r;
+  ctrl[C_BETA] = beta;
+  ctrl[C_EPD] = beta * (ctrl[C_EPD] + alpha * ctrl[C_DPD]);
+  ctrl[C_DPD] = z_r_new + beta * beta * ctrl[C_DPD];
+  ctrl[C_ZR] = z_r_new;
+  ctrl[C_ITER] += 1.0;
+}
+
+// end of the unrolled loop: cap at max_inner
+__global__ void k_ctrl_tcg_end(double* ctrl) {
+  if (threadIdx.x != 0) return;
+  if (ctrl[C_STATUS] != (double)ST_RUN) return;
+  ctrl[C_STATUS] = (double)ST_TCG_STOP;
+  ctrl[C_J] = ctrl[C_ITER];
+  ctrl[C_HLEN] = ctrl[C_ITER];
+  ctrl[C_USE_CURRENT] = 1.0;
+}
+
+// candidate selection for the current radius: replay the stored Krylov
+// scalars, find the truncation point and the model decrease.
+__global__ void k_ctrl_candidate(double* ctrl) {
+  if (threadIdx.x != 0) return;
+  if (ctrl[C_STATUS] != (double)ST_TCG_STOP) return;
+  const double radius = ctrl[C_RADIUS];
+  const int hlen = (int)ctrl[C_HLEN];
+  double dm = 0.0;
+  int jstar = -1;
+  double taustar = 0.0;
+  for (int j = 0; j < hlen && j < MAX_TCG; ++j) {
+    const double z_r = ctrl[H_ZR(j)];
+    const double d_Hd = ctrl[H_DHD(j)];
+    const double e_Pe = ctrl[H_EPE(j)];
+    const double e_Pd = ctrl[H_EPD(j)];
+    const double d_Pd = ctrl[H_DPD(j)];
+    const double alpha = z_r / d_Hd;
+    const double e_Pe_new = e_Pe + 2.0 * alpha * e_Pd + alpha * alpha * d_Pd;
+    if (d_Hd <= 0.0 || e_Pe_new >= radius * radius) {
+      const double disc = e_Pd * e_Pd + d_Pd * (radius * radius - e_Pe);
+      const double tau = (d_Pd > 0.0)
+          ? (-e_Pd + sqrt(fmax(disc, 0.0))) / d_Pd : 0.0;
+      dm += tau * z_r - 0.5 * tau * tau * d_Hd;
+      jstar = j;
+      taustar = tau;
+      break;
+    }
+    dm += 0.5 * alpha * z_r;
+  }
+  if (jstar < 0) {
+    // full step (no truncation at this radius)
+    ctrl[C_USE_CURRENT] = 1.0;
+  } else {
+    ctrl[C_USE_CURRENT] = 0.0;
+    ctrl[C_JSTAR] = (double)jstar;
+    ctrl[C_TAUSTAR] = taustar;
+  }
+  ctrl[C_DM] = dm;
+  ctrl[C_DOT0] = 0.0;
+  ctrl[C_DOT2] = 0.0;
+}
+
+// acceptance test after f(X_prop) dots (C_DOT0 = <QXp, Xp>, C_DOT2 = <G, Xp>)
+__global__ void k_ctrl_accept(double* ctrl, double accept_rho) {
+  if (threadIdx.x != 0) return;
+  if (ctrl[C_STATUS] != (double)ST_TCG_STOP) return;
+  const double fprop = 0.5 * ctrl[C_DOT0] + ctrl[C_DOT2];
+  ctrl[C_DOT0] = 0.0;
+  ctrl[C_DOT2] = 0.0;
+  ctrl[C_FPROP] = fprop;
+  const double fX = ctrl[C_FX];
+  const double dm = ctrl[C_DM];
+  const double rho = (fX - fprop) / fmax(dm, 1e-300);
+  ctrl[C_RHO] = rho;
+  if (rho > accept_rho && fprop <= fX) {
+    ctrl[C_STATUS] = (double)ST_ACCEPTED;
+  }
+}
+
+// shrink the radius after a rejection (host re-enqueues the candidate set)
+__global__ void k_ctrl_shrink(double* ctrl) {
+  if (threadIdx.x != 0) return;
+  if (ctrl[C_STATUS] != (double)ST_TCG_STOP) return;
+  ctrl[C_RADIUS] *= 0.25;
+}
+
+// ---------------------------------------------------------------------
+// Assembly kernels
+// ---------------------------------------------------------------------
+// Q values: vals[slot] += w[edge_of[c]] * blocks[c]  (scatter-add)
+__global__ void k_q_assemble(double* __restrict__ vals,
+                             const double* __restrict__ blocks,
+                             const long* __restrict__ slots,
+                             const long* __restrict__ edge_of,
+                             const double* __restrict__ w,
+                             int ncontrib, int bsz) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= (long)ncontrib * bsz) return;
+  const int c = i / bsz, e = i % bsz;
+  atomicAdd(&vals[slots[c] * bsz + e], w[edge_of[c]] * blocks[i]);
+}
+
+// G values: Gt[local_pose[e]] += -w[e] * E0[e] @ nbr[slot[e]]
+// one thread per (edge, c, k) output element
+__global__ void k_g_assemble(double* __restrict__ Gt,
+                             const double* __restrict__ E0,
+                             const long* __restrict__ local_pose,
+                             const long* __restrict__ nbr_slot,
+                             const double* __restrict__ nbr,
+                             const double* __restrict__ w,
+                             int ne, int dh, int r) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= (long)ne * dh * r) return;
+  const int e = i / (dh * r);
+  const int t = i % (dh * r);
+  const int c = t / r, k = t % r;
+  const double* Ee = E0 + (size_t)e * dh * dh;
+  const double* Xn = nbr + (size_t)nbr_slot[e] * dh * r;
+  double acc = 0.0;
+  for (int b = 0; b < dh; ++b) acc = fma(Ee[c * dh + b], Xn[b * r + k], acc);
+  atomicAdd(&Gt[((size_t)local_pose[e] * dh + c) * r + k], -w[e] * acc);
+}
+
+// ---------------------------------------------------------------------
+// C ABI
+// ---------------------------------------------------------------------
+static inline int blocks_for(long total, int bs) {
+  return (int)((total + bs - 1) / bs);
+}
+
+extern "C" {
+
+void dpo_bsr_spmm(const int* row_ptr, const int* col_idx, const double* vals,
+                  int n, int dh, const double* X, double* out, int r,
+                  const double* ctrl, int guard, void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  const int tile = dh * r;
+  const int per_block = 256 / tile;
+  const int grid = blocks_for(n, per_block);
+  hipLaunchKernelGGL(k_bsr_spmm, dim3(grid), dim3(256), 0, s,
+                     row_ptr, col_idx, vals, X, out, n, dh, r, ctrl, guard);
+}
+
+void dpo_proj_dots(const double* X, const double* V, const double* G,
+                   double* out, const double* dotWith, double* ctrl,
+                   int n, int d, int r, int dot_slot, int dot_slot2,
+                   int neg, int guard, void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  const int grid = blocks_for(n, 256);
+  if (neg)
+    hipLaunchKernelGGL(k_proj_dots<1>, dim3(grid), dim3(256), 0, s,
+                       X, V, G, out, dotWith, ctrl, n, d, r,
+                       dot_slot, dot_slot2, guard);
+  else
+    hipLaunchKernelGGL(k_proj_dots<0>, dim3(grid), dim3(256), 0, s,
+                       X, V, G, out, dotWith, ctrl, n, d, r,
+                       dot_slot, dot_slot2, guard);
+}
+
+void dpo_polar_affine(const double* A, const double* B, const double* C,
+                      double ca, double cb, double cc, double* out,
+                      int n, int d, int r, const double* ctrl, int guard,
+                      void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  const int grid = blocks_for(n, 256);
+  hipLaunchKernelGGL(k_polar_affine, dim3(grid), dim3(256), 0, s,
+                     A, B, C, ca, cb, cc, out, n, d, r, ctrl, guard);
+}
+
+void dpo_precond_dense(const float* Minv, const double* V, double* Z,
+                       int N, int r, const double* ctrl, int guard,
+                       void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  const int grid = blocks_for(N, 256);
+  const size_t shmem = 64 * r * sizeof(double);
+  hipLaunchKernelGGL(k_precond_dense, dim3(grid), dim3(256), shmem, s,
+                     Minv, V, Z, N, r, ctrl, guard);
+}
+
+void dpo_precond_jacobi(const double* L, const double* V, double* Z,
+                        int n, int dh, int r, const double* ctrl, int guard,
+                        void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  const int grid = blocks_for((long)n * r, 256);
+  hipLaunchKernelGGL(k_precond_jacobi, dim3(grid), dim3(256), 0, s,
+                     L, V, Z, n, dh, r, ctrl, guard);
+}
+
+void dpo_tcg_update(double* eta, double* rvec, const double* delta,
+                    const double* Hd, double* eta_snap, double* delta_snap,
+                    double* ctrl, long total, void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  hipLaunchKernelGGL(k_tcg_update, dim3(blocks_for(total, 256)), dim3(256),
+                     0, s, eta, rvec, delta, Hd, eta_snap, delta_snap,
+                     ctrl, total);
+}
+
+void dpo_tcg_delta(double* delta, const double* z, double* ctrl, long total,
+                   void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  hipLaunchKernelGGL(k_tcg_delta, dim3(blocks_for(total, 256)), dim3(256),
+                     0, s, delta, z, ctrl, total);
+}
+
+void dpo_form_step(double* step, const double* eta, const double* eta_snap,
+                   const double* delta_snap, const double* ctrl, long total,
+                   void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  hipLaunchKernelGGL(k_form_step, dim3(blocks_for(total, 256)), dim3(256),
+                     0, s, step, eta, eta_snap, delta_snap, ctrl, total);
+}
+
+void dpo_axpby(const double* A, const double* B, double a, double b,
+               double* out, long total, void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  hipLaunchKernelGGL(k_axpby, dim3(blocks_for(total, 256)), dim3(256),
+                     0, s, A, B, a, b, out, total);
+}
+
+void dpo_dots(const double* A, const double* B, const double* C,
+              double* ctrl, int slot, int slot2, long total, int guard,
+              void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  hipLaunchKernelGGL(k_dots, dim3(blocks_for(total, 256)), dim3(256),
+                     0, s, A, B, C, ctrl, slot, slot2, total, guard);
+}
+
+void dpo_ctrl_init(double* ctrl, double tol, double Delta0, double theta,
+                   double kappa, void* stream) {
+  hipLaunchKernelGGL(k_ctrl_init, dim3(1), dim3(64), 0,
+                     (hipStream_t)stream, ctrl, tol, Delta0, theta, kappa);
+}
+void dpo_ctrl_z0(double* ctrl, void* stream) {
+  hipLaunchKernelGGL(k_ctrl_z0, dim3(1), dim3(64), 0, (hipStream_t)stream,
+                     ctrl);
+}
+void dpo_ctrl_alpha(double* ctrl, void* stream) {
+  hipLaunchKernelGGL(k_ctrl_alpha, dim3(1), dim3(64), 0,
+                     (hipStream_t)stream, ctrl);
+}
+void dpo_ctrl_rr(double* ctrl, void* stream) {
+  hipLaunchKernelGGL(k_ctrl_rr, dim3(1), dim3(64), 0, (hipStream_t)stream,
+                     ctrl);
+}
+void dpo_ctrl_beta(double* ctrl, void* stream) {
+  hipLaunchKernelGGL(k_ctrl_beta, dim3(1), dim3(64), 0, (hipStream_t)stream,
+                     ctrl);
+}
+void dpo_ctrl_tcg_end(double* ctrl, void* stream) {
+  hipLaunchKernelGGL(k_ctrl_tcg_end, dim3(1), dim3(64), 0,
+                     (hipStream_t)stream, ctrl);
+}
+void dpo_ctrl_candidate(double* ctrl, void* stream) {
+  hipLaunchKernelGGL(k_ctrl_candidate, dim3(1), dim3(64), 0,
+                     (hipStream_t)stream, ctrl);
+}
+void dpo_ctrl_accept(double* ctrl, double accept_rho, void* stream) {
+  hipLaunchKernelGGL(k_ctrl_accept, dim3(1), dim3(64), 0,
+                     (hipStream_t)stream, ctrl, accept_rho);
+}
+void dpo_ctrl_shrink(double* ctrl, void* stream) {
+  hipLaunchKernelGGL(k_ctrl_shrink, dim3(1), dim3(64), 0,
+                     (hipStream_t)stream, ctrl);
+}
+
+void dpo_q_assemble(double* vals, const double* blocks, const long* slots,
+                    const long* edge_of, const double* w, int ncontrib,
+                    int bsz, long nnzb, void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  DPO_CHECK(hipMemsetAsync(vals, 0, (size_t)nnzb * bsz * sizeof(double), s));
+  hipLaunchKernelGGL(k_q_assemble,
+                     dim3(blocks_for((long)ncontrib * bsz, 256)), dim3(256),
+                     0, s, vals, blocks, slots, edge_of, w, ncontrib, bsz);
+}
+
+void dpo_g_assemble(double* Gt, const double* E0, const long* local_pose,
+                    const long* nbr_slot, const double* nbr, const double* w,
+                    int ne, int dh, int r, long N, void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  DPO_CHECK(hipMemsetAsync(Gt, 0, (size_t)N * r * sizeof(double), s));
+  hipLaunchKernelGGL(k_g_assemble,
+                     dim3(blocks_for((long)ne * dh * r, 256)), dim3(256),
+                     0, s, Gt, E0, local_pose, nbr_slot, nbr, w, ne, dh, r);
+}
+
+int dpo_ctrl_size() { return CTRL_SIZE; }
+
+}  // extern "C"

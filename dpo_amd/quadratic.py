@@ -131,7 +131,11 @@ class BSRMatrix:
         return self.to_scalar_csr().to_dense().to(self.vals.device)
 
     def spmm(self, X: Tensor) -> Tensor:
-        """Q @ X (CPU reference via scalar CSR; the HIP path overrides)."""
+        """Q @ X — HIP BSR kernel on GPU, scalar-CSR torch spmm on CPU."""
+        if X.is_cuda:
+            from .ops import hip_backend
+            return hip_backend.bsr_spmm(self.row_ptr, self.col_idx,
+                                        self.vals, self.n, self.dh, X)
         return torch.sparse.mm(self.to_scalar_csr(), X)
 
     def to(self, device) -> "BSRMatrix":
@@ -279,16 +283,31 @@ class GAssembler:
 
     def assemble(self, nbr_poses: Tensor, weights: Tensor, r: int) -> Tensor:
         """nbr_poses: (n_slots, dh, r) packed neighbor poses (Xt blocks);
-        weights: (ne,) per-shared-edge GNC weights. Returns Gt (N, r)."""
+        weights: (ne,) per-shared-edge GNC weights. Returns Gt (N, r).
+        GPU path: one HIP scatter-add kernel (rebuilt every iteration —
+        SURVEY.md 2c row 'G assembly')."""
         dh = self.d + 1
         dev = nbr_poses.device
-        E0 = self.E0.to(dev)
-        w = weights.to(dev)
-        Xn = nbr_poses[self.nbr_slot.to(dev)]        # (ne, dh, r)
+        if dev.type != "cpu":
+            from .ops import hip_backend
+            if getattr(self, "_dev_cache", None) is None or \
+                    self._dev_cache[0] != dev:
+                self._dev_cache = (dev, self.E0.to(dev).contiguous(),
+                                   self.local_pose.to(dev),
+                                   self.nbr_slot.to(dev))
+            _, E0, lp, slots = self._dev_cache
+            Gt = torch.empty(self.n * dh, r, dtype=torch.float64, device=dev)
+            hip_backend.g_assemble(Gt, E0, lp, slots,
+                                   nbr_poses.contiguous(),
+                                   weights.to(dev).contiguous(), dh, r)
+            return Gt
+        E0 = self.E0
+        w = weights
+        Xn = nbr_poses[self.nbr_slot]                # (ne, dh, r)
         contrib = -torch.bmm(E0 * w[:, None, None], Xn)
-        Gt = torch.zeros(self.n * dh, r, dtype=torch.float64, device=dev)
+        Gt = torch.zeros(self.n * dh, r, dtype=torch.float64)
         Gb = Gt.view(self.n, dh, r)
-        Gb.index_add_(0, self.local_pose.to(dev), contrib)
+        Gb.index_add_(0, self.local_pose, contrib)
         return Gt
 
 
@@ -348,7 +367,7 @@ class QuadraticProblem:
             A = Q.to_dense()
             A += precond_reg * torch.eye(self.N, dtype=A.dtype, device=dev)
             L = torch.linalg.cholesky(A)
-            self._Minv = torch.cholesky_inverse(L).to(torch.float32)
+            self._Minv = torch.cholesky_inverse(L).to(torch.float32).contiguous()
         else:
             diag = Q.diag_blocks() + precond_reg * torch.eye(
                 dh, dtype=torch.float64, device=dev)
