@@ -233,96 +233,59 @@ __global__ void k_proj_dots(const double* __restrict__ X,
 // Translations pass through the affine combination unchanged.
 // One thread per pose, everything in registers.
 // ---------------------------------------------------------------------
-__device__ void sym_eig(const double S[3][3], int d, double lam[3],
-                        double Q[3][3]) {
-  if (d == 1) {
-    lam[0] = S[0][0];
-    Q[0][0] = 1.0;
+// Inverse square root of a d x d SPD Gram matrix via the coupled
+// Newton-Schulz iteration (trace-scaled):  Y -> A^{1/2}, Z -> A^{-1/2}.
+// Branch-free and exact in the fully-degenerate case (Gram ~ c*I), which
+// is the common case for retractions (X + small step is near-Stiefel) —
+// an analytic eigenvector decomposition is ill-conditioned exactly there.
+__device__ void spd_inv_sqrt(const double S[3][3], int d, double out[3][3]) {
+  double tr = 0.0;
+  for (int i = 0; i < d; ++i) tr += S[i][i];
+  if (tr <= 1e-300) {
+    for (int i = 0; i < d; ++i)
+      for (int j = 0; j < d; ++j) out[i][j] = (i == j) ? 0.0 : 0.0;
     return;
   }
-  if (d == 2) {
-    double tr = S[0][0] + S[1][1];
-    double det = S[0][0] * S[1][1] - S[0][1] * S[1][0];
-    double disc = sqrt(fmax(tr * tr * 0.25 - det, 0.0));
-    lam[0] = tr * 0.5 + disc;
-    lam[1] = tr * 0.5 - disc;
-    // eigenvector for lam[0]
-    double a = S[0][0] - lam[1], b = S[0][1];
-    double n0 = hypot(a, S[1][0]);
-    if (n0 > 1e-300) {
-      Q[0][0] = a / n0; Q[1][0] = S[1][0] / n0;
-    } else {
-      Q[0][0] = 1.0; Q[1][0] = 0.0;
+  const double inv_s = 1.0 / tr;
+  double Y[3][3], Z[3][3];
+  for (int i = 0; i < d; ++i)
+    for (int j = 0; j < d; ++j) {
+      Y[i][j] = S[i][j] * inv_s;
+      Z[i][j] = (i == j) ? 1.0 : 0.0;
     }
-    Q[0][1] = -Q[1][0];
-    Q[1][1] = Q[0][0];
-    (void)b;
-    return;
-  }
-  // d == 3: trigonometric (Smith) method + one inverse-iteration polish.
-  double p1 = S[0][1] * S[0][1] + S[0][2] * S[0][2] + S[1][2] * S[1][2];
-  double q = (S[0][0] + S[1][1] + S[2][2]) / 3.0;
-  double a00 = S[0][0] - q, a11 = S[1][1] - q, a22 = S[2][2] - q;
-  double p2 = a00 * a00 + a11 * a11 + a22 * a22 + 2.0 * p1;
-  double p = sqrt(fmax(p2 / 6.0, 0.0));
-  if (p < 1e-300) {
-    lam[0] = lam[1] = lam[2] = q;
-    for (int i = 0; i < 3; ++i)
-      for (int j = 0; j < 3; ++j) Q[i][j] = (i == j) ? 1.0 : 0.0;
-    return;
-  }
-  double invp = 1.0 / p;
-  // B = (S - q I) / p ; r = det(B)/2
-  double B[3][3];
-  B[0][0] = a00 * invp; B[0][1] = S[0][1] * invp; B[0][2] = S[0][2] * invp;
-  B[1][0] = B[0][1];    B[1][1] = a11 * invp;     B[1][2] = S[1][2] * invp;
-  B[2][0] = B[0][2];    B[2][1] = B[1][2];        B[2][2] = a22 * invp;
-  double detB = B[0][0] * (B[1][1] * B[2][2] - B[1][2] * B[2][1])
-              - B[0][1] * (B[1][0] * B[2][2] - B[1][2] * B[2][0])
-              + B[0][2] * (B[1][0] * B[2][1] - B[1][1] * B[2][0]);
-  double rr = fmin(fmax(detB * 0.5, -1.0), 1.0);
-  double phi = acos(rr) / 3.0;
-  lam[0] = q + 2.0 * p * cos(phi);
-  lam[2] = q + 2.0 * p * cos(phi + 2.0943951023931953);  // + 2pi/3
-  lam[1] = 3.0 * q - lam[0] - lam[2];
-  // eigenvectors: v_k = (S - lam_a I)(S - lam_b I) e_col, pick best col
-  for (int k = 0; k < 3; ++k) {
-    double la = lam[(k + 1) % 3], lb = lam[(k + 2) % 3];
-    double M[3][3];
-    for (int i = 0; i < 3; ++i)
-      for (int j = 0; j < 3; ++j) {
+  for (int it = 0; it < 40; ++it) {
+    // T = (3 I - Z Y) / 2
+    double T[3][3];
+    double delta = 0.0;
+    for (int i = 0; i < d; ++i)
+      for (int j = 0; j < d; ++j) {
         double acc = 0.0;
-        for (int m = 0; m < 3; ++m) {
-          double s_im = S[i][m] - (i == m ? la : 0.0);
-          double s_mj = S[m][j] - (m == j ? lb : 0.0);
-          acc += s_im * s_mj;
+        for (int k = 0; k < d; ++k) acc = fma(Z[i][k], Y[k][j], acc);
+        T[i][j] = 0.5 * (((i == j) ? 3.0 : 0.0) - acc);
+        const double dij = T[i][j] - ((i == j) ? 1.0 : 0.0);
+        delta += dij * dij;
+      }
+    double Yn[3][3], Zn[3][3];
+    for (int i = 0; i < d; ++i)
+      for (int j = 0; j < d; ++j) {
+        double ay = 0.0, az = 0.0;
+        for (int k = 0; k < d; ++k) {
+          ay = fma(Y[i][k], T[k][j], ay);
+          az = fma(T[i][k], Z[k][j], az);
         }
-        M[i][j] = acc;
+        Yn[i][j] = ay;
+        Zn[i][j] = az;
       }
-    // column with largest norm
-    int bc = 0; double bn = -1.0;
-    for (int j = 0; j < 3; ++j) {
-      double nn = M[0][j] * M[0][j] + M[1][j] * M[1][j] + M[2][j] * M[2][j];
-      if (nn > bn) { bn = nn; bc = j; }
-    }
-    double nv = sqrt(fmax(bn, 1e-300));
-    Q[0][k] = M[0][bc] / nv;
-    Q[1][k] = M[1][bc] / nv;
-    Q[2][k] = M[2][bc] / nv;
+    for (int i = 0; i < d; ++i)
+      for (int j = 0; j < d; ++j) {
+        Y[i][j] = Yn[i][j];
+        Z[i][j] = Zn[i][j];
+      }
+    if (delta < 1e-32) break;  // converged (T ~ I)
   }
-  // re-orthogonalize (degenerate eigenvalues): Gram-Schmidt
-  for (int k = 1; k < 3; ++k)
-    for (int m = 0; m < k; ++m) {
-      double dp = Q[0][k] * Q[0][m] + Q[1][k] * Q[1][m] + Q[2][k] * Q[2][m];
-      Q[0][k] -= dp * Q[0][m];
-      Q[1][k] -= dp * Q[1][m];
-      Q[2][k] -= dp * Q[2][m];
-      double nn = sqrt(Q[0][k] * Q[0][k] + Q[1][k] * Q[1][k]
-                       + Q[2][k] * Q[2][k]);
-      if (nn > 1e-300) {
-        Q[0][k] /= nn; Q[1][k] /= nn; Q[2][k] /= nn;
-      }
-    }
+  const double c = rsqrt(tr);
+  for (int i = 0; i < d; ++i)
+    for (int j = 0; j < d; ++j) out[i][j] = Z[i][j] * c;
 }
 
 __global__ void k_polar_affine(const double* __restrict__ A,
@@ -355,20 +318,8 @@ __global__ void k_polar_affine(const double* __restrict__ A,
       for (int k = 0; k < r; ++k) s = fma(Mt[a][k], Mt[b][k], s);
       S[a][b] = s;
     }
-  double lam[3], Q[3][3];
-  sym_eig(S, d, lam, Q);
-  // G^{-1/2} = Q diag(lam^{-1/2}) Q^T with clamping
-  double lmax = fmax(lam[0], 1e-300);
   double Gi[3][3];
-  for (int a = 0; a < d; ++a)
-    for (int b = 0; b < d; ++b) {
-      double s = 0.0;
-      for (int k = 0; k < d; ++k) {
-        double l = fmax(lam[k], 1e-14 * lmax);
-        s += Q[a][k] * Q[b][k] * rsqrt(l);
-      }
-      Gi[a][b] = s;
-    }
+  spd_inv_sqrt(S, d, Gi);
   double* Oi = out + (size_t)i * dh * r;
   for (int a = 0; a < d; ++a)
     for (int k = 0; k < r; ++k) {
