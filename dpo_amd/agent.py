@@ -766,6 +766,75 @@ class PGOAgent:
         self.X = None
 
     # ------------------------------------------------------------------
+    # packed fast path (GPU): device-tensor neighbor buffers + native
+    # C++ solve/eval, driven by DistributedRBCDDriver._run_packed.
+    # ------------------------------------------------------------------
+    def _ensure_packed(self, dev) -> None:
+        if getattr(self, "_packed_ready", False):
+            return
+        n_slots = len(self._nbr_slot_order)
+        self._nbr_buffer = torch.zeros(max(n_slots, 1), self.dh, self.r,
+                                       dtype=torch.float64, device=dev)
+        self._nbr_buffer_aux = torch.zeros_like(self._nbr_buffer)
+        self._w_shared_dev = torch.tensor(
+            [m.weight for m in self.shared_lc], dtype=torch.float64,
+            device=dev)
+        if getattr(self, "_dev_solver", None) is None:
+            from .ops.hip_backend import DeviceSolver
+            self._dev_solver = DeviceSolver(self.n, self.d, self.r, dev,
+                                            max_inner=10)
+        if self.params.acceleration:
+            self.Y = self.X.clone()
+            self.V = self.X.clone()
+            self.gamma = 0.0
+            self.alpha = 0.0
+        self._packed_ready = True
+
+    def _packed_set_g(self, aux: bool) -> None:
+        if self.shared_lc:
+            buf = self._nbr_buffer_aux if aux else self._nbr_buffer
+            Gt = self._g_assembler.assemble(buf, self._w_shared_dev, self.r)
+            self.problem.set_g(Gt)
+        else:
+            self.problem.set_g(None)
+
+    def _packed_solve(self, accel: bool) -> None:
+        self._packed_set_g(aux=accel)
+        if accel:
+            self.X.copy_(self.Y)
+        self._dev_solver.solve(self.problem, self.X, tol=1e-2,
+                               Delta0=100.0, compute_final_gradnorm=False)
+
+    def _packed_eval(self):
+        """Device 3-vector [f, 0.5<X,G>, ||rgrad||^2] with fresh G."""
+        self._packed_set_g(aux=False)
+        return self._dev_solver.eval_terms(self.problem, self.X)
+
+    def _packed_nesterov_pre(self) -> None:
+        from .ops import hip_backend as hb
+        self._update_gamma()
+        self._update_alpha()
+        self._XPrev_packed = self.X.clone()
+        self.Y = hb.polar_affine(self.X, self.V, None,
+                                 1.0 - self.alpha, self.alpha, 0.0, self.d)
+
+    def _packed_nesterov_post(self, it: int) -> None:
+        from .ops import hip_backend as hb
+        self.V = hb.polar_affine(self.V, self.X, self.Y,
+                                 1.0, self.gamma, -self.gamma, self.d)
+        if (self.iteration_number + it + 1) % self.params.restart_interval == 0:
+            # periodic restart (PGOAgent.cpp:1040-1052)
+            self.X.copy_(self._XPrev_packed)
+            self._packed_set_g(aux=False)
+            self._dev_solver.solve(self.problem, self.X, tol=1e-2,
+                                   Delta0=100.0,
+                                   compute_final_gradnorm=False)
+            self.V.copy_(self.X)
+            self.Y.copy_(self.X)
+            self.gamma = 0.0
+            self.alpha = 0.0
+
+    # ------------------------------------------------------------------
     # asynchronous optimization loop (PGOAgent.cpp:861-916)
     # ------------------------------------------------------------------
     def start_optimization_loop(self, rate_hz: float) -> None:

@@ -147,6 +147,10 @@ class DistributedRBCDDriver:
         self._colors = colors
         self._num_colors = max(colors) + 1 if colors else 1
 
+    def _robust_is_l2(self) -> bool:
+        return all(a.params.robust_cost_type == RobustCostType.L2
+                   for a in self.local_agents.values())
+
     @staticmethod
     def _public_pose_sets(shared_lc, num_robots):
         pub = [set() for _ in range(num_robots)]
@@ -260,6 +264,13 @@ class DistributedRBCDDriver:
     def run(self, max_iters: int = 1000, gradnorm_tol: float = 0.1,
             trace_file: Optional[str] = None,
             time_limit_s: Optional[float] = None) -> RBCDResult:
+        import torch
+        if (str(self.device).startswith("cuda")
+                and self._robust_is_l2()
+                and all(a.state == PGOAgentState.INITIALIZED
+                        for a in self.local_agents.values())):
+            return self._run_packed(max_iters, gradnorm_tol, trace_file,
+                                    time_limit_s)
         res = RBCDResult()
         selected = 0
         t0 = time.perf_counter()
@@ -303,3 +314,171 @@ class DistributedRBCDDriver:
             fout.close()
         res.elapsed_s = time.perf_counter() - t0
         return res
+
+    # ===================================================================
+    # Packed fast path (GPU, L2 cost, all agents initialized):
+    # device-tensor payloads end-to-end, native C++ solve + eval, one
+    # host sync per round. Used automatically by run() when applicable.
+    # ===================================================================
+    def _packed_setup(self):
+        if getattr(self, "_pk", None) is not None:
+            return
+        import torch
+        dev = torch.device(self.device)
+        blk = self.dh * self.r
+        pk = {}
+        # per-agent public index tensors & rank payload layout
+        pk["pub_idx_t"] = {}
+        offs = {}
+        off = 0
+        for rb in self.rank_agents[self.comm.rank]:
+            offs[rb] = off
+            off += len(self.pub_idx[rb]) * blk
+        pk["rank_payload_len"] = [
+            sum(len(self.pub_idx[rb]) * blk for rb in agents)
+            for agents in self.rank_agents]
+        pk["local_off"] = offs
+        # global offsets: agent rb's block inside its owner's payload
+        glob_off = {}
+        for rk, agents in enumerate(self.rank_agents):
+            o = 0
+            for rb in agents:
+                glob_off[rb] = o
+                o += len(self.pub_idx[rb]) * blk
+        pk["glob_off"] = glob_off
+        # per local agent: for each neighbor agent, (src positions within
+        # neighbor pub list, dst slots in local nbr buffer)
+        pk["scatter"] = {}
+        for rb, a in self.local_agents.items():
+            a._ensure_packed(dev)
+            pk["pub_idx_t"][rb] = torch.tensor(
+                self.pub_idx[rb], dtype=torch.int64, device=dev)
+            per_nbr = {}
+            pos_in_pub = {nb: {p: k for k, p in enumerate(self.pub_idx[nb])}
+                          for nb in a.get_neighbors()}
+            for slot, (nb, p) in enumerate(a._nbr_slot_order):
+                per_nbr.setdefault(nb, ([], []))
+                per_nbr[nb][0].append(pos_in_pub[nb][p])
+                per_nbr[nb][1].append(slot)
+            pk["scatter"][rb] = {
+                nb: (torch.tensor(src, dtype=torch.int64, device=dev),
+                     torch.tensor(dst, dtype=torch.int64, device=dev))
+                for nb, (src, dst) in per_nbr.items()}
+        pk["dev"] = dev
+        self._pk = pk
+
+    def _packed_pack(self, use_aux: bool = False):
+        import torch
+        pk = self._pk
+        blk = self.dh * self.r
+        parts = []
+        for rb in self.rank_agents[self.comm.rank]:
+            a = self.local_agents[rb]
+            src = a.Y if (use_aux and a.Y is not None) else a.X
+            Xb = src.view(a.n, self.dh, self.r)
+            parts.append(Xb.index_select(0, pk["pub_idx_t"][rb]).reshape(-1))
+        if parts:
+            return torch.cat(parts)
+        return torch.zeros(0, dtype=torch.float64, device=pk["dev"])
+
+    def _packed_scatter(self, flats, aux: bool = False):
+        pk = self._pk
+        blk = self.dh * self.r
+        for rb, a in self.local_agents.items():
+            for nb, (src_idx, dst_slots) in pk["scatter"][rb].items():
+                payload = flats[self.owner[nb]]
+                o = pk["glob_off"][nb]
+                npub = len(self.pub_idx[nb])
+                blkv = payload[o:o + npub * blk].view(npub, self.dh, self.r)
+                buf = a._nbr_buffer_aux if aux else a._nbr_buffer
+                buf.index_copy_(0, dst_slots, blkv.index_select(0, src_idx))
+
+    def _run_packed(self, max_iters, gradnorm_tol, trace_file, time_limit_s):
+        import torch
+        from .types import OptAlgorithm
+        res = RBCDResult()
+        self._packed_setup()
+        pk = self._pk
+        dev = pk["dev"]
+        selected = 0
+        t0 = time.perf_counter()
+        evalmat = torch.zeros(self.num_robots, 3, dtype=torch.float64,
+                              device=dev)
+        sizes = pk["rank_payload_len"]
+        accel = self.acceleration
+        # Nesterov host-side scalars per agent
+        if accel:
+            for a in self.local_agents.values():
+                a.gamma = 0.0
+                a.alpha = 0.0
+        flats = self.comm.all_gather_flat(self._packed_pack(), sizes)
+        self._packed_scatter(flats)
+        fout = open(trace_file, "w") if (trace_file and
+                                         self.comm.rank == 0) else None
+        for it in range(max_iters):
+            if self.selection == "colored":
+                color = it % self._num_colors
+                active = [rb for rb in range(self.num_robots)
+                          if self._colors[rb] == color]
+            else:
+                active = [selected]
+            if accel:
+                # Y-update first, then exchange aux poses so the active
+                # agents solve against same-round Y (reference pulls aux
+                # dicts right before the selected iterate,
+                # MultiRobotExample.cpp:259-273).
+                for a in self.local_agents.values():
+                    a._packed_nesterov_pre()
+                aux_flats = self.comm.all_gather_flat(
+                    self._packed_pack(use_aux=True), sizes)
+                self._packed_scatter(aux_flats, aux=True)
+            for rb, a in self.local_agents.items():
+                if rb in active:
+                    a._packed_solve(accel)
+                elif accel:
+                    a.X.copy_(a.Y)
+            if accel:
+                for a in self.local_agents.values():
+                    a._packed_nesterov_post(it)
+            flats = self.comm.all_gather_flat(self._packed_pack(), sizes)
+            self._packed_scatter(flats)
+            # evaluation (fresh neighbor data)
+            evalmat.zero_()
+            for rb, a in self.local_agents.items():
+                evalmat[rb] = a._packed_eval()
+            self.comm.all_reduce_sum_(evalmat)
+            ev = evalmat.cpu().numpy()          # the round's one host sync
+            cost = float((ev[:, 0] - ev[:, 1]).sum())
+            gn2 = ev[:, 2]
+            gradnorm = float(np.sqrt(gn2.sum()))
+            res.trace.append((2.0 * cost, gradnorm))
+            if fout:
+                fout.write(f"{2.0 * cost:.10g},{gradnorm:.10g}\n")
+            res.iterations = it + 1
+            if gradnorm < gradnorm_tol:
+                res.converged = True
+                break
+            if time_limit_s and time.perf_counter() - t0 > time_limit_s:
+                break
+            selected = int(np.argmax(gn2))
+        if res.trace:
+            res.final_cost, res.final_gradnorm = res.trace[-1]
+        if fout:
+            fout.close()
+        res.elapsed_s = time.perf_counter() - t0
+        # sync dict-path state (global anchor for rounding)
+        self._sync_anchor()
+        return res
+
+    def _sync_anchor(self):
+        import torch
+        blk = self.dh * self.r
+        dev = self._pk["dev"] if getattr(self, "_pk", None) else "cpu"
+        buf = torch.zeros(blk, dtype=torch.float64, device=dev)
+        if 0 in self.local_agents:
+            a0 = self.local_agents[0]
+            buf.copy_(a0.X.view(a0.n, self.dh, self.r)[0].reshape(-1))
+        self.comm.all_reduce_sum_(buf)
+        M = buf.cpu().numpy().reshape(self.dh, self.r).T
+        for a in self.local_agents.values():
+            a.set_global_anchor(np.ascontiguousarray(M))

@@ -48,6 +48,14 @@ _lib.dpo_ctrl_accept.argtypes = [_c, _d, _c]
 _lib.dpo_q_assemble.argtypes = [_c, _c, _c, _c, _c, _i, _i, _l, _c]
 _lib.dpo_g_assemble.argtypes = [_c, _c, _c, _c, _c, _c, _i, _i, _i, _l, _c]
 _lib.dpo_ctrl_size.restype = _i
+_lib.dpo_ctx_create.restype = _c
+_lib.dpo_ctx_create.argtypes = [_i, _i, _i, _i]
+_lib.dpo_ctx_destroy.argtypes = [_c]
+_lib.dpo_ctx_set_problem.argtypes = [_c, _c, _c, _c, _c, _c, _c]
+_lib.dpo_rbcd_solve.restype = _i
+_lib.dpo_rbcd_solve.argtypes = [_c, _c, _d, _d, _i, _d, _i,
+                                ctypes.POINTER(ctypes.c_double), _c]
+_lib.dpo_eval_terms.argtypes = [_c, _c, _c, _c]
 
 CTRL_SIZE = _lib.dpo_ctrl_size()
 
@@ -180,154 +188,68 @@ def q_assemble(vals: Tensor, blocks: Tensor, slots: Tensor, edge_of: Tensor,
 # Device-resident RBCD local solve
 # ---------------------------------------------------------------------
 class DeviceSolver:
-    """Workspace + orchestration for the trust-region RBCD local solve on
-    one agent's problem, entirely on-device.
+    """Native (C++) RBCD trust-region local solve with device-resident
+    tCG control. Semantics mirror reference QuadraticOptimizer::
+    trustRegion with Max_Iteration == 1 (QuadraticOptimizer.cpp:92-110):
+    one Steihaug-tCG TR step, radius shrunk /4 until accepted (<= 10
+    shrinks). The Krylov path is radius-independent, so rejections replay
+    stored snapshots instead of re-running tCG. One host sync per solve
+    in the accepted-first-try case; ~1 enqueue call from Python."""
 
-    Semantics mirror reference QuadraticOptimizer::trustRegion with
-    Max_Iteration == 1 (QuadraticOptimizer.cpp:92-110): one Steihaug-tCG
-    trust-region step, shrinking the radius /4 until accepted (<= 10
-    shrinks). The Krylov path is radius-independent, so the shrink loop
-    REPLAYS the stored per-iteration scalars + snapshots instead of
-    re-running tCG — each rejection costs one retraction + one f eval.
-    """
-
-    def __init__(self, n: int, d: int, r: int, device,
-                 max_inner: int = 10):
+    def __init__(self, n: int, d: int, r: int, device, max_inner: int = 10):
         assert max_inner <= MAX_TCG
         self.n, self.d, self.r = n, d, r
         self.dh = d + 1
         self.N = self.dh * n
         self.max_inner = max_inner
-        dev = torch.device(device)
-        total = self.N * r
-        f64 = dict(dtype=torch.float64, device=dev)
-        self.W = torch.empty(self.N, r, **f64)
-        self.grad = torch.empty(self.N, r, **f64)
-        self.eta = torch.empty(self.N, r, **f64)
-        self.delta = torch.empty(self.N, r, **f64)
-        self.rvec = torch.empty(self.N, r, **f64)
-        self.z = torch.empty(self.N, r, **f64)
-        self.Hd = torch.empty(self.N, r, **f64)
-        self.step = torch.empty(self.N, r, **f64)
-        self.Xprop = torch.empty(self.N, r, **f64)
-        self.eta_snap = torch.empty(max_inner + 1, self.N, r, **f64)
-        self.delta_snap = torch.empty(max_inner + 1, self.N, r, **f64)
-        self.ctrl = torch.zeros(CTRL_SIZE, **f64)
-        self.total = total
+        self.device = torch.device(device)
+        with torch.cuda.device(self.device):
+            self.handle = _lib.dpo_ctx_create(n, d, r, max_inner)
+        self._stats = (ctypes.c_double * 8)()
+        self._eval_out = torch.zeros(3, dtype=torch.float64,
+                                     device=self.device)
+        self._refs = None  # keep problem tensors alive across the call
 
-    def _precond(self, problem, V, out):
-        ctrl, s = self.ctrl, _stream(V)
-        if problem._Minv is not None:
-            _lib.dpo_precond_dense(_p(problem._Minv), _p(V), _p(out),
-                                   self.N, self.r, _p(ctrl), GUARD_RUN, s)
-        else:
-            _lib.dpo_precond_jacobi(_p(problem._Lpre), _p(V), _p(out),
-                                    self.n, self.dh, self.r, _p(ctrl),
-                                    GUARD_RUN, s)
+    def __del__(self):
+        try:
+            if getattr(self, "handle", None):
+                _lib.dpo_ctx_destroy(self.handle)
+        except Exception:
+            pass
+
+    def _bind(self, problem):
+        Q = problem.Q
+        G = problem.Gt
+        Minv = getattr(problem, "_Minv", None)
+        Ljac = getattr(problem, "_Lpre", None)
+        if Minv is None and Ljac is not None:
+            Ljac = Ljac.contiguous()
+        self._refs = (Q.row_ptr, Q.col_idx, Q.vals, G, Minv, Ljac)
+        _lib.dpo_ctx_set_problem(self.handle, _p(Q.row_ptr), _p(Q.col_idx),
+                                 _p(Q.vals), _p(G), _p(Minv), _p(Ljac))
 
     def solve(self, problem, X: Tensor, tol: float = 1e-2,
               Delta0: float = 100.0, max_shrink: int = 10,
-              accept_rho: float = 0.1, theta: float = 1.0,
-              kappa: float = 0.1, compute_final_gradnorm: bool = True):
-        """Run one RBCD local solve in place on X. Returns a stats dict
-        (f_init, grad_norm_init, f_opt, grad_norm_opt, status)."""
-        Q = problem.Q
-        G = problem.Gt
-        d, r, n, N = self.d, self.r, self.n, self.N
-        ctrl = self.ctrl
-        s = _stream(X)
-        rp, ci, vals = Q.row_ptr, Q.col_idx, Q.vals
-
-        ctrl.zero_()
-        self.eta.zero_()
-        self.delta.zero_()
-        # --- gradient phase ------------------------------------------
-        _lib.dpo_bsr_spmm(_p(rp), _p(ci), _p(vals), n, self.dh, _p(X),
-                          _p(self.W), r, None, GUARD_NONE, s)
-        # grad = P_X(W + G); C_DOT1 = ||grad||^2 ; C_DOT0 = <W+G, X>
-        _lib.dpo_proj_dots(_p(X), _p(self.W), _p(G), _p(self.grad), None,
-                           _p(ctrl), n, d, r, C_DOT1, C_DOT0, 0,
-                           GUARD_NONE, s)
-        if G is not None:
-            _lib.dpo_dots(_p(G), _p(X), None, _p(ctrl), C_DOT2, -1,
-                          self.total, GUARD_NONE, s)
-        # r0 = grad
-        _lib.dpo_axpby(_p(self.grad), None, 1.0, 0.0, _p(self.rvec),
-                       self.total, s)
-        _lib.dpo_ctrl_init(_p(ctrl), tol, Delta0, theta, kappa, s)
-        # z0 = P_X(M^-1 r0); C_DOT0 = <z0, r0>
-        self._precond(problem, self.rvec, self.z)
-        _lib.dpo_proj_dots(_p(X), _p(self.z), None, _p(self.z), _p(self.rvec),
-                           _p(ctrl), n, d, r, C_DOT0, -1, 0, GUARD_RUN, s)
-        _lib.dpo_ctrl_z0(_p(ctrl), s)
-        # delta0 = -z (beta = 0)
-        _lib.dpo_tcg_delta(_p(self.delta), _p(self.z), _p(ctrl), self.total, s)
-
-        # --- tCG loop (guarded, unrolled) ----------------------------
-        for _ in range(self.max_inner):
-            _lib.dpo_bsr_spmm(_p(rp), _p(ci), _p(vals), n, self.dh,
-                              _p(self.delta), _p(self.Hd), r, _p(ctrl),
-                              GUARD_RUN, s)
-            _lib.dpo_proj_dots(_p(X), _p(self.Hd), None, _p(self.Hd),
-                               _p(self.delta), _p(ctrl), n, d, r,
-                               C_DOT0, -1, 0, GUARD_RUN, s)
-            _lib.dpo_ctrl_alpha(_p(ctrl), s)
-            _lib.dpo_tcg_update(_p(self.eta), _p(self.rvec), _p(self.delta),
-                                _p(self.Hd), _p(self.eta_snap),
-                                _p(self.delta_snap), _p(ctrl), self.total, s)
-            _lib.dpo_ctrl_rr(_p(ctrl), s)
-            self._precond(problem, self.rvec, self.z)
-            _lib.dpo_proj_dots(_p(X), _p(self.z), None, _p(self.z),
-                               _p(self.rvec), _p(ctrl), n, d, r,
-                               C_DOT0, -1, 0, GUARD_RUN, s)
-            _lib.dpo_ctrl_beta(_p(ctrl), s)
-            _lib.dpo_tcg_delta(_p(self.delta), _p(self.z), _p(ctrl),
-                               self.total, s)
-        _lib.dpo_ctrl_tcg_end(_p(ctrl), s)
-
-        # --- candidate / shrink loop ---------------------------------
-        status = ST_GIVE_UP
-        for attempt in range(max_shrink + 1):
-            _lib.dpo_ctrl_candidate(_p(ctrl), s)
-            _lib.dpo_form_step(_p(self.step), _p(self.eta), _p(self.eta_snap),
-                               _p(self.delta_snap), _p(ctrl), self.total, s)
-            _lib.dpo_polar_affine(_p(X), _p(self.step), None, 1.0, 1.0, 0.0,
-                                  _p(self.Xprop), n, d, r, _p(ctrl),
-                                  GUARD_STOP, s)
-            _lib.dpo_bsr_spmm(_p(rp), _p(ci), _p(vals), n, self.dh,
-                              _p(self.Xprop), _p(self.W), r, _p(ctrl),
-                              GUARD_STOP, s)
-            _lib.dpo_dots(_p(self.Xprop), _p(self.W), _p(G), _p(ctrl),
-                          C_DOT0, C_DOT2, self.total, GUARD_STOP, s)
-            _lib.dpo_ctrl_accept(_p(ctrl), accept_rho, s)
-            st = int(ctrl[C_STATUS].item())  # host sync
-            if st == ST_ACCEPTED:
-                X.copy_(self.Xprop)
-                status = st
-                break
-            if st == ST_NO_UPDATE:
-                status = st
-                break
-            _lib.dpo_ctrl_shrink(_p(ctrl), s)
-        # --- stats ----------------------------------------------------
-        stats_t = self.ctrl.cpu()
-        stats = {
-            "status": status,
-            "f_init": float(stats_t[C_FX]),
-            "grad_norm_init": float(stats_t[C_GN0SQ]) ** 0.5,
-            "f_opt": float(stats_t[C_FPROP]) if status == ST_ACCEPTED
-            else float(stats_t[C_FX]),
-            "rho": float(stats_t[C_RHO]),
+              accept_rho: float = 0.1, compute_final_gradnorm: bool = True):
+        """One RBCD local solve in place on X. Returns a stats dict."""
+        self._bind(problem)
+        status = _lib.dpo_rbcd_solve(
+            self.handle, _p(X), tol, Delta0, max_shrink, accept_rho,
+            1 if compute_final_gradnorm else 0, self._stats, _stream(X))
+        st = self._stats
+        return {
+            "status": int(st[0]),
+            "f_init": st[1],
+            "grad_norm_init": st[2],
+            "f_opt": st[3],
+            "grad_norm_opt": st[4],
+            "rho": st[5],
+            "shrinks": int(st[6]),
         }
-        if compute_final_gradnorm and status == ST_ACCEPTED:
-            ctrl[C_DOT0] = 0.0
-            ctrl[C_DOT1] = 0.0
-            _lib.dpo_bsr_spmm(_p(rp), _p(ci), _p(vals), n, self.dh, _p(X),
-                              _p(self.W), r, None, GUARD_NONE, s)
-            _lib.dpo_proj_dots(_p(X), _p(self.W), _p(G), _p(self.grad),
-                               None, _p(ctrl), n, d, r, C_DOT1, -1, 0,
-                               GUARD_NONE, s)
-            stats["grad_norm_opt"] = float(ctrl[C_DOT1].item()) ** 0.5
-        elif status in (ST_NO_UPDATE, ST_GIVE_UP):
-            stats["grad_norm_opt"] = stats["grad_norm_init"]
-        return stats
+
+    def eval_terms(self, problem, X: Tensor) -> Tensor:
+        """Device 3-vector [f(X), 0.5<X,G>, ||rgrad||^2]; no host sync."""
+        self._bind(problem)
+        _lib.dpo_eval_terms(self.handle, _p(X), _p(self._eval_out),
+                            _stream(X))
+        return self._eval_out

@@ -178,3 +178,35 @@ def test_driver_gpu_converges():
                            device=DEV)
     res = drv.run(max_iters=300)
     assert res.converged, f"gradnorm {res.final_gradnorm}"
+
+
+def test_dist_driver_packed_gpu_matches_cpu():
+    """DistributedRBCDDriver world=1: GPU packed fast path vs CPU dict
+    path must produce matching cost traces and converge."""
+    from dpo_amd.comm import Comm
+    from dpo_amd.dist_driver import DistributedRBCDDriver
+    from dpo_amd.synthetic import grid3d
+    meas, n = grid3d(side=4, seed=0)
+    cpu = DistributedRBCDDriver(meas, n, 2, Comm(), r=5,
+                                partition="contiguous")
+    res_cpu = cpu.run(max_iters=200)
+    gpu = DistributedRBCDDriver(meas, n, 2, Comm(), r=5,
+                                partition="contiguous", device=DEV)
+    res_gpu = gpu.run(max_iters=200)
+    assert res_gpu.converged
+    for (c0, _), (c1, _) in zip(res_cpu.trace[:20], res_gpu.trace[:20]):
+        assert abs(c0 - c1) < 1e-4 * max(1.0, abs(c0))
+    assert abs(res_gpu.final_cost - res_cpu.final_cost) < 1e-3 * max(
+        1.0, abs(res_cpu.final_cost))
+
+
+def test_dist_driver_packed_gpu_accelerated():
+    from dpo_amd.comm import Comm
+    from dpo_amd.dist_driver import DistributedRBCDDriver
+    from dpo_amd.synthetic import grid3d
+    meas, n = grid3d(side=4, seed=0)
+    gpu = DistributedRBCDDriver(meas, n, 2, Comm(), r=5,
+                                partition="contiguous", device=DEV,
+                                acceleration=True)
+    res = gpu.run(max_iters=250)
+    assert res.converged
