@@ -65,12 +65,19 @@ def _sparse_lsq(A: sp.spmatrix, b: np.ndarray) -> np.ndarray:
     """min ||A x + b||_2 via normal equations with a small Tikhonov-free
     sparse factorization (the systems here are full-rank by construction
     once the first pose is pinned)."""
+    import warnings
     AtA = (A.T @ A).tocsc()
     Atb = A.T @ b
-    try:
-        x = spla.spsolve(AtA, -Atb)
-    except RuntimeError:
-        x = spla.lsqr(A, -b, atol=1e-12, btol=1e-12, iter_lim=10000)[0]
+    with warnings.catch_warnings():
+        warnings.simplefilter("ignore")
+        try:
+            x = spla.spsolve(AtA, -Atb)
+        except RuntimeError:
+            x = None
+    if x is None or not np.all(np.isfinite(x)):
+        # rank-deficient (e.g. a disconnected local subgraph): min-norm
+        # least-squares via LSQR
+        x = spla.lsqr(A, -b, atol=1e-12, btol=1e-12, iter_lim=20000)[0]
     return np.asarray(x)
 
 

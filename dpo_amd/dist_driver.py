@@ -82,6 +82,8 @@ class DistributedRBCDDriver:
         self.local_agents: Dict[int, PGOAgent] = {}
         from .manifold import lifting_matrix
         YL = lifting_matrix(d, r)
+        T_chordal_pre = chordal_initialization(d, num_poses, measurements) \
+            if robust == RobustCostType.L2 else None
         for rb in range(num_robots):
             if self.owner[rb] != comm.rank:
                 continue
@@ -91,7 +93,15 @@ class DistributedRBCDDriver:
                                verbose=verbose, device=device)
             a = PGOAgent(rb, p)
             a.set_lifting_matrix(YL)
-            a.set_pose_graph(odometry[rb], private_lc[rb], shared_lc[rb])
+            T_init = None
+            if T_chordal_pre is not None:
+                T_init = np.zeros((d, self.pose_counts[rb] * self.dh))
+                for i in range(self.pose_counts[rb]):
+                    g = self.pose_to_index[(rb, i)]
+                    T_init[:, i * self.dh:(i + 1) * self.dh] = \
+                        T_chordal_pre[:, g * self.dh:(g + 1) * self.dh]
+            a.set_pose_graph(odometry[rb], private_lc[rb], shared_lc[rb],
+                             T_init=T_init)
             self.local_agents[rb] = a
 
         # ---- public-pose packing layout (global, same on all ranks) ----
@@ -119,8 +129,7 @@ class DistributedRBCDDriver:
             for agents in self.rank_agents]
 
         # ---- centralized chordal init (identical on all ranks) ---------
-        T_chordal = chordal_initialization(d, num_poses, measurements) \
-            if robust == RobustCostType.L2 else None
+        T_chordal = T_chordal_pre
         if T_chordal is not None:
             X_chordal = YL @ T_chordal
             for rb, a in self.local_agents.items():

@@ -92,6 +92,24 @@ class MultiRobotDriver:
             Qc = Qc.to(device)
         self.central.set_q(Qc)
 
+        # ---- centralized chordal init, sliced per agent ----------------
+        # (computed before agents so set_pose_graph can take it as TInit
+        # and skip redundant per-agent local initializations)
+        # The reference example initializes from the centralized chordal
+        # relaxation regardless of the robust mode
+        # (MultiRobotExample.cpp:185-202).
+        T_chordal = chordal_initialization(d, num_poses, measurements)
+
+        def _slice_T(rb):
+            if T_chordal is None:
+                return None
+            Tr = np.zeros((d, self.pose_counts[rb] * self.dh))
+            for i in range(self.pose_counts[rb]):
+                g = self.pose_to_index[(rb, i)]
+                Tr[:, i * self.dh:(i + 1) * self.dh] = \
+                    T_chordal[:, g * self.dh:(g + 1) * self.dh]
+            return Tr
+
         # ---- agents ----------------------------------------------------
         self.agents: List[PGOAgent] = []
         for rb in range(num_robots):
@@ -105,11 +123,10 @@ class MultiRobotDriver:
             a = PGOAgent(rb, p)
             if rb > 0:
                 a.set_lifting_matrix(self.agents[0].get_lifting_matrix())
-            a.set_pose_graph(odometry[rb], private_lc[rb], shared_lc[rb])
+            a.set_pose_graph(odometry[rb], private_lc[rb], shared_lc[rb],
+                             T_init=_slice_T(rb))
             self.agents.append(a)
 
-        # ---- centralized chordal init, distributed via set_x ------------
-        T_chordal = chordal_initialization(d, num_poses, measurements)
         YL = self.agents[0].get_lifting_matrix()
         X_chordal = YL @ T_chordal  # (r, (d+1) n)
         for rb in range(num_robots):
