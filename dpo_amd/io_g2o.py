@@ -137,3 +137,33 @@ def adjacency_from_measurements(
         nbr[m.p1].add(m.p2)
         nbr[m.p2].add(m.p1)
     return [sorted(s) for s in nbr]
+
+
+def load_npz_dataset(path: str) -> Tuple[List[RelativeSEMeasurement], int]:
+    """Load a dataset converted to .npz by scripts/convert_datasets
+    (same measurement content as the .g2o original)."""
+    z = np.load(path)
+    d = int(z["d"])
+    n = int(z["n"])
+    R, t = z["R"], z["t"]
+    p1, p2 = z["p1"], z["p2"]
+    kappa, tau = z["kappa"], z["tau"]
+    meas = [RelativeSEMeasurement(0, 0, int(p1[k]), int(p2[k]),
+                                  R[k].copy(), t[k].copy(),
+                                  float(kappa[k]), float(tau[k]))
+            for k in range(len(p1))]
+    return meas, n
+
+
+def load_dataset(name_or_path: str) -> Tuple[List[RelativeSEMeasurement], int]:
+    """Load by dataset name (data/<name>.npz), .npz path or .g2o path."""
+    import os
+    if name_or_path.endswith(".g2o"):
+        return read_g2o(name_or_path)
+    if name_or_path.endswith(".npz"):
+        return load_npz_dataset(name_or_path)
+    here = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    p = os.path.join(here, "data", name_or_path + ".npz")
+    if os.path.exists(p):
+        return load_npz_dataset(p)
+    raise FileNotFoundError(f"dataset {name_or_path} not found")
