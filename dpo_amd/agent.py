@@ -783,6 +783,13 @@ class PGOAgent:
             from .ops.hip_backend import DeviceSolver
             self._dev_solver = DeviceSolver(self.n, self.d, self.r, dev,
                                             max_inner=10)
+        ga = self._g_assembler
+        if getattr(ga, "_dev_cache", None) is None or ga._dev_cache[0] != dev:
+            ga._dev_cache = (dev, ga.E0.to(dev).contiguous(),
+                             ga.local_pose.to(dev), ga.nbr_slot.to(dev))
+        _, E0, lp, slots = ga._dev_cache
+        self._dev_solver.set_gdata(E0, lp, slots, self._w_shared_dev)
+        self._dev_solver.bind_problem_static(self.problem)
         if self.params.acceleration:
             self.Y = self.X.clone()
             self.V = self.X.clone()
@@ -799,16 +806,14 @@ class PGOAgent:
             self.problem.set_g(None)
 
     def _packed_solve(self, accel: bool) -> None:
-        self._packed_set_g(aux=accel)
         if accel:
             self.X.copy_(self.Y)
-        self._dev_solver.solve(self.problem, self.X, tol=1e-2,
-                               Delta0=100.0, compute_final_gradnorm=False)
+        nbr = self._nbr_buffer_aux if accel else self._nbr_buffer
+        self._dev_solver.round_solve(self.X, nbr, tol=1e-2, Delta0=100.0)
 
     def _packed_eval(self):
         """Device 3-vector [f, 0.5<X,G>, ||rgrad||^2] with fresh G."""
-        self._packed_set_g(aux=False)
-        return self._dev_solver.eval_terms(self.problem, self.X)
+        return self._dev_solver.round_eval(self.X, self._nbr_buffer)
 
     def _packed_nesterov_pre(self) -> None:
         from .ops import hip_backend as hb
@@ -825,10 +830,8 @@ class PGOAgent:
         if (self.iteration_number + it + 1) % self.params.restart_interval == 0:
             # periodic restart (PGOAgent.cpp:1040-1052)
             self.X.copy_(self._XPrev_packed)
-            self._packed_set_g(aux=False)
-            self._dev_solver.solve(self.problem, self.X, tol=1e-2,
-                                   Delta0=100.0,
-                                   compute_final_gradnorm=False)
+            self._dev_solver.round_solve(self.X, self._nbr_buffer,
+                                         tol=1e-2, Delta0=100.0)
             self.V.copy_(self.X)
             self.Y.copy_(self.X)
             self.gamma = 0.0

@@ -905,7 +905,23 @@ struct DpoCtx {
   const double *Gt = nullptr;
   const float *Minv = nullptr;
   const double *Ljac = nullptr;
+  // G-assembly structure (owned by torch tensors)
+  const double *g_E0 = nullptr;
+  const long *g_local_pose = nullptr, *g_nbr_slot = nullptr;
+  const double *g_w = nullptr;
+  int g_ne = 0;
+  double *G_buf = nullptr;  // ctx-owned (N, r) linear-term buffer
 };
+
+static void ctx_assemble_g(DpoCtx* c, const double* nbr, hipStream_t s) {
+  if (c->g_ne == 0) { c->Gt = nullptr; return; }
+  DPO_CHECK(hipMemsetAsync(c->G_buf, 0, c->total * sizeof(double), s));
+  hipLaunchKernelGGL(k_g_assemble,
+                     dim3(((long)c->g_ne * c->dh * c->r + 255) / 256),
+                     dim3(256), 0, s, c->G_buf, c->g_E0, c->g_local_pose,
+                     c->g_nbr_slot, nbr, c->g_w, c->g_ne, c->dh, c->r);
+  c->Gt = c->G_buf;
+}
 
 static void ctx_precond(DpoCtx* c, const double* V, double* Z,
                         hipStream_t s) {
@@ -951,6 +967,7 @@ void* dpo_ctx_create(int n, int d, int r, int max_inner) {
   DPO_CHECK(hipMalloc(&c->eta_snap, vb * (max_inner + 1)));
   DPO_CHECK(hipMalloc(&c->delta_snap, vb * (max_inner + 1)));
   DPO_CHECK(hipMalloc(&c->ctrl, CTRL_SIZE * sizeof(double)));
+  DPO_CHECK(hipMalloc(&c->G_buf, vb));
   DPO_CHECK(hipHostMalloc(&c->ctrl_host, CTRL_SIZE * sizeof(double)));
   return c;
 }
@@ -960,7 +977,7 @@ void dpo_ctx_destroy(void* h) {
   hipFree(c->W); hipFree(c->grad); hipFree(c->eta); hipFree(c->delta);
   hipFree(c->rvec); hipFree(c->z); hipFree(c->Hd); hipFree(c->step);
   hipFree(c->Xprop); hipFree(c->eta_snap); hipFree(c->delta_snap);
-  hipFree(c->ctrl); hipHostFree(c->ctrl_host);
+  hipFree(c->ctrl); hipFree(c->G_buf); hipHostFree(c->ctrl_host);
   delete c;
 }
 
@@ -1113,6 +1130,34 @@ void dpo_eval_terms(void* h, const double* X, double* out_dev,
                        C_DOT2, -1, total, -1);
   hipLaunchKernelGGL(k_eval_combine, dim3(1), dim3(64), 0, s, c->ctrl,
                      out_dev);
+}
+
+
+// Install the static G-assembly structure (per-agent, set once; weights
+// pointer re-read every assembly so GNC re-weighting is free).
+void dpo_ctx_set_gdata(void* h, const double* E0, const long* local_pose,
+                       const long* nbr_slot, const double* w, int ne) {
+  DpoCtx* c = (DpoCtx*)h;
+  c->g_E0 = E0; c->g_local_pose = local_pose; c->g_nbr_slot = nbr_slot;
+  c->g_w = w; c->g_ne = ne;
+}
+
+// Fused per-round entry points: assemble G from the packed neighbor
+// buffer, then solve / evaluate. One Python call each.
+int dpo_round_solve(void* h, double* X, const double* nbr, double tol,
+                    double Delta0, int max_shrink, double accept_rho,
+                    double* stats_out, void* stream) {
+  DpoCtx* c = (DpoCtx*)h;
+  ctx_assemble_g(c, nbr, (hipStream_t)stream);
+  return dpo_rbcd_solve(h, X, tol, Delta0, max_shrink, accept_rho, 0,
+                        stats_out, stream);
+}
+
+void dpo_round_eval(void* h, const double* X, const double* nbr,
+                    double* out_dev, void* stream) {
+  DpoCtx* c = (DpoCtx*)h;
+  ctx_assemble_g(c, nbr, (hipStream_t)stream);
+  dpo_eval_terms(h, X, out_dev, stream);
 }
 
 }  // extern "C"

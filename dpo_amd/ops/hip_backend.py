@@ -56,6 +56,11 @@ _lib.dpo_rbcd_solve.restype = _i
 _lib.dpo_rbcd_solve.argtypes = [_c, _c, _d, _d, _i, _d, _i,
                                 ctypes.POINTER(ctypes.c_double), _c]
 _lib.dpo_eval_terms.argtypes = [_c, _c, _c, _c]
+_lib.dpo_ctx_set_gdata.argtypes = [_c, _c, _c, _c, _c, _i]
+_lib.dpo_round_solve.restype = _i
+_lib.dpo_round_solve.argtypes = [_c, _c, _c, _d, _d, _i, _d,
+                                 ctypes.POINTER(ctypes.c_double), _c]
+_lib.dpo_round_eval.argtypes = [_c, _c, _c, _c, _c]
 
 CTRL_SIZE = _lib.dpo_ctrl_size()
 
@@ -252,4 +257,34 @@ class DeviceSolver:
         self._bind(problem)
         _lib.dpo_eval_terms(self.handle, _p(X), _p(self._eval_out),
                             _stream(X))
+        return self._eval_out
+
+    # --- fused per-round entry points (G assembled in C++) -----------
+    def set_gdata(self, E0: Tensor, local_pose: Tensor, nbr_slot: Tensor,
+                  w: Tensor) -> None:
+        self._grefs = (E0, local_pose, nbr_slot, w)
+        _lib.dpo_ctx_set_gdata(self.handle, _p(E0), _p(local_pose),
+                               _p(nbr_slot), _p(w), E0.shape[0])
+
+    def bind_problem_static(self, problem) -> None:
+        """Bind the Q/preconditioner pointers once (re-call after a Q
+        rebuild). The G term is produced in-C++ by the round calls."""
+        Q = problem.Q
+        Minv = getattr(problem, "_Minv", None)
+        Ljac = getattr(problem, "_Lpre", None)
+        if Minv is None and Ljac is not None:
+            Ljac = Ljac.contiguous()
+        self._refs = (Q.row_ptr, Q.col_idx, Q.vals, None, Minv, Ljac)
+        _lib.dpo_ctx_set_problem(self.handle, _p(Q.row_ptr), _p(Q.col_idx),
+                                 _p(Q.vals), None, _p(Minv), _p(Ljac))
+
+    def round_solve(self, X: Tensor, nbr: Tensor, tol: float = 1e-2,
+                    Delta0: float = 100.0) -> int:
+        return _lib.dpo_round_solve(self.handle, _p(X), _p(nbr), tol,
+                                    Delta0, 10, 0.1, self._stats,
+                                    _stream(X))
+
+    def round_eval(self, X: Tensor, nbr: Tensor) -> Tensor:
+        _lib.dpo_round_eval(self.handle, _p(X), _p(nbr),
+                            _p(self._eval_out), _stream(X))
         return self._eval_out
