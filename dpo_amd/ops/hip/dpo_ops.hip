@@ -115,13 +115,44 @@ __device__ __forceinline__ void block_reduce_atomic(double v, double* dst) {
 // blocks. Q block values are broadcast from L2 (tiny per-agent Q is
 // cache-resident); X block reads are contiguous dh*r segments.
 // ---------------------------------------------------------------------
+template <int DH, int R>
 __global__ void k_bsr_spmm(const int* __restrict__ row_ptr,
                            const int* __restrict__ col_idx,
                            const double* __restrict__ vals,
                            const double* __restrict__ X,
                            double* __restrict__ out,
-                           int n, int dh, int r,
-                           const double* __restrict__ ctrl, int guard) {
+                           int n, const double* __restrict__ ctrl,
+                           int guard) {
+  if (guarded_off(ctrl, guard)) return;
+  constexpr int tile = DH * R;
+  constexpr int per_block = 256 / tile;
+  const int slot = threadIdx.x / tile;
+  const int t = threadIdx.x % tile;
+  const int i = blockIdx.x * per_block + slot;
+  if (threadIdx.x >= per_block * tile || i >= n) return;
+  const int c = t / R;   // row inside block
+  const int k = t % R;   // column of X
+  const int s = row_ptr[i], e = row_ptr[i + 1];
+  double acc = 0.0;
+  for (int p = s; p < e; ++p) {
+    const int j = col_idx[p];
+    const double* B = vals + (size_t)p * DH * DH;
+    const double* Xj = X + (size_t)j * DH * R;
+    #pragma unroll
+    for (int cc = 0; cc < DH; ++cc)
+      acc = fma(B[c * DH + cc], Xj[cc * R + k], acc);
+  }
+  out[(size_t)i * DH * R + c * R + k] = acc;
+}
+
+// generic-(dh, r) fallback (rare shapes); compile-time tile kernels above
+__global__ void k_bsr_spmm_gen(const int* __restrict__ row_ptr,
+                               const int* __restrict__ col_idx,
+                               const double* __restrict__ vals,
+                               const double* __restrict__ X,
+                               double* __restrict__ out,
+                               int n, int dh, int r,
+                               const double* __restrict__ ctrl, int guard) {
   if (guarded_off(ctrl, guard)) return;
   const int tile = dh * r;
   const int per_block = blockDim.x / tile;
@@ -129,15 +160,14 @@ __global__ void k_bsr_spmm(const int* __restrict__ row_ptr,
   const int t = threadIdx.x % tile;
   const int i = blockIdx.x * per_block + slot;
   if (threadIdx.x >= per_block * tile || i >= n) return;
-  const int c = t / r;   // row inside block
-  const int k = t % r;   // column of X
+  const int c = t / r;
+  const int k = t % r;
   const int s = row_ptr[i], e = row_ptr[i + 1];
   double acc = 0.0;
   for (int p = s; p < e; ++p) {
     const int j = col_idx[p];
     const double* B = vals + (size_t)p * dh * dh;
     const double* Xj = X + (size_t)j * dh * r;
-    #pragma unroll 4
     for (int cc = 0; cc < dh; ++cc)
       acc = fma(B[c * dh + cc], Xj[cc * r + k], acc);
   }
@@ -155,68 +185,81 @@ __global__ void k_bsr_spmm(const int* __restrict__ row_ptr,
 #define DPO_MAX_R 8
 #define DPO_MAX_DH 4
 
-template <int NEG>
+template <int NEG, int D, int R>
 __global__ void k_proj_dots(const double* __restrict__ X,
                             const double* __restrict__ V,
                             const double* __restrict__ G,
                             double* __restrict__ out,
                             const double* __restrict__ dotWith,
                             double* __restrict__ ctrl,
-                            int n, int d, int r,
-                            int dot_slot, int dot_slot2,
+                            int n, int dot_slot, int dot_slot2,
                             int guard) {
   if (guarded_off(ctrl, guard)) return;
   const int i = blockIdx.x * blockDim.x + threadIdx.x;
-  const int dh = d + 1;
-  double Yt[DPO_MAX_DH - 1][DPO_MAX_R];
-  double Vt[DPO_MAX_DH][DPO_MAX_R];
+  constexpr int dh = D + 1;
+  double Yt[D][R];
+  double Vt[dh][R];
   double dot_vx = 0.0;
   if (i < n) {
-    const double* Xi = X + (size_t)i * dh * r;
-    const double* Vi = V + (size_t)i * dh * r;
-    const double* Gi = G ? G + (size_t)i * dh * r : nullptr;
+    const double* Xi = X + (size_t)i * dh * R;
+    const double* Vi = V + (size_t)i * dh * R;
+    const double* Gi = G ? G + (size_t)i * dh * R : nullptr;
+    #pragma unroll
     for (int c = 0; c < dh; ++c)
-      for (int k = 0; k < r; ++k) {
-        double v = Vi[c * r + k];
-        if (Gi) v += Gi[c * r + k];
+      #pragma unroll
+      for (int k = 0; k < R; ++k) {
+        double v = Vi[c * R + k];
+        if (Gi) v += Gi[c * R + k];
         Vt[c][k] = v;
-        if (dot_slot2 >= 0) dot_vx = fma(v, Xi[c * r + k], dot_vx);
+        if (dot_slot2 >= 0) dot_vx = fma(v, Xi[c * R + k], dot_vx);
       }
-    for (int c = 0; c < d; ++c)
-      for (int k = 0; k < r; ++k)
-        Yt[c][k] = Xi[c * r + k];
-    // S = sym(Yt Vt^T)  (d x d)
-    double S[DPO_MAX_DH - 1][DPO_MAX_DH - 1];
-    for (int a = 0; a < d; ++a)
-      for (int b = 0; b < d; ++b) {
+    #pragma unroll
+    for (int c = 0; c < D; ++c)
+      #pragma unroll
+      for (int k = 0; k < R; ++k)
+        Yt[c][k] = Xi[c * R + k];
+    // S = sym(Yt Vt^T)  (D x D)
+    double S[D][D];
+    #pragma unroll
+    for (int a = 0; a < D; ++a)
+      #pragma unroll
+      for (int b = 0; b < D; ++b) {
         double s = 0.0;
-        for (int k = 0; k < r; ++k) s = fma(Yt[a][k], Vt[b][k], s);
+        #pragma unroll
+        for (int k = 0; k < R; ++k) s = fma(Yt[a][k], Vt[b][k], s);
         S[a][b] = s;
       }
-    for (int a = 0; a < d; ++a)
-      for (int b = a; b < d; ++b) {
+    #pragma unroll
+    for (int a = 0; a < D; ++a)
+      #pragma unroll
+      for (int b = a; b < D; ++b) {
         double s = 0.5 * (S[a][b] + S[b][a]);
         S[a][b] = s;
         S[b][a] = s;
       }
     // P = Vt - S Yt on Stiefel rows
-    for (int a = 0; a < d; ++a)
-      for (int k = 0; k < r; ++k) {
+    #pragma unroll
+    for (int a = 0; a < D; ++a)
+      #pragma unroll
+      for (int k = 0; k < R; ++k) {
         double acc = Vt[a][k];
-        for (int b = 0; b < d; ++b) acc = fma(-S[a][b], Yt[b][k], acc);
+        #pragma unroll
+        for (int b = 0; b < D; ++b) acc = fma(-S[a][b], Yt[b][k], acc);
         Vt[a][k] = acc;
       }
   }
   double dot_pw = 0.0;
   if (i < n) {
-    double* Oi = out + (size_t)i * dh * r;
-    const double* Wi = dotWith ? dotWith + (size_t)i * dh * r : nullptr;
+    double* Oi = out + (size_t)i * dh * R;
+    const double* Wi = dotWith ? dotWith + (size_t)i * dh * R : nullptr;
+    #pragma unroll
     for (int c = 0; c < dh; ++c)
-      for (int k = 0; k < r; ++k) {
+      #pragma unroll
+      for (int k = 0; k < R; ++k) {
         double p = NEG ? -Vt[c][k] : Vt[c][k];
-        Oi[c * r + k] = p;
+        Oi[c * R + k] = p;
         if (dot_slot >= 0) {
-          double w = Wi ? Wi[c * r + k] : Vt[c][k];  // default: <P,P>
+          double w = Wi ? Wi[c * R + k] : Vt[c][k];  // default: <P,P>
           dot_pw = fma(p, w, dot_pw);
         }
       }
@@ -238,96 +281,123 @@ __global__ void k_proj_dots(const double* __restrict__ X,
 // Branch-free and exact in the fully-degenerate case (Gram ~ c*I), which
 // is the common case for retractions (X + small step is near-Stiefel) —
 // an analytic eigenvector decomposition is ill-conditioned exactly there.
-__device__ void spd_inv_sqrt(const double S[3][3], int d, double out[3][3]) {
+template <int D>
+__device__ __forceinline__ void spd_inv_sqrt(const double S[D][D],
+                                             double out[D][D]) {
   double tr = 0.0;
-  for (int i = 0; i < d; ++i) tr += S[i][i];
+  #pragma unroll
+  for (int i = 0; i < D; ++i) tr += S[i][i];
   if (tr <= 1e-300) {
-    for (int i = 0; i < d; ++i)
-      for (int j = 0; j < d; ++j) out[i][j] = (i == j) ? 0.0 : 0.0;
+    #pragma unroll
+    for (int i = 0; i < D; ++i)
+      #pragma unroll
+      for (int j = 0; j < D; ++j) out[i][j] = 0.0;
     return;
   }
   const double inv_s = 1.0 / tr;
-  double Y[3][3], Z[3][3];
-  for (int i = 0; i < d; ++i)
-    for (int j = 0; j < d; ++j) {
+  double Y[D][D], Z[D][D];
+  #pragma unroll
+  for (int i = 0; i < D; ++i)
+    #pragma unroll
+    for (int j = 0; j < D; ++j) {
       Y[i][j] = S[i][j] * inv_s;
       Z[i][j] = (i == j) ? 1.0 : 0.0;
     }
   for (int it = 0; it < 40; ++it) {
     // T = (3 I - Z Y) / 2
-    double T[3][3];
+    double T[D][D];
     double delta = 0.0;
-    for (int i = 0; i < d; ++i)
-      for (int j = 0; j < d; ++j) {
+    #pragma unroll
+    for (int i = 0; i < D; ++i)
+      #pragma unroll
+      for (int j = 0; j < D; ++j) {
         double acc = 0.0;
-        for (int k = 0; k < d; ++k) acc = fma(Z[i][k], Y[k][j], acc);
+        #pragma unroll
+        for (int k = 0; k < D; ++k) acc = fma(Z[i][k], Y[k][j], acc);
         T[i][j] = 0.5 * (((i == j) ? 3.0 : 0.0) - acc);
         const double dij = T[i][j] - ((i == j) ? 1.0 : 0.0);
         delta += dij * dij;
       }
-    double Yn[3][3], Zn[3][3];
-    for (int i = 0; i < d; ++i)
-      for (int j = 0; j < d; ++j) {
+    double Yn[D][D], Zn[D][D];
+    #pragma unroll
+    for (int i = 0; i < D; ++i)
+      #pragma unroll
+      for (int j = 0; j < D; ++j) {
         double ay = 0.0, az = 0.0;
-        for (int k = 0; k < d; ++k) {
+        #pragma unroll
+        for (int k = 0; k < D; ++k) {
           ay = fma(Y[i][k], T[k][j], ay);
           az = fma(T[i][k], Z[k][j], az);
         }
         Yn[i][j] = ay;
         Zn[i][j] = az;
       }
-    for (int i = 0; i < d; ++i)
-      for (int j = 0; j < d; ++j) {
+    #pragma unroll
+    for (int i = 0; i < D; ++i)
+      #pragma unroll
+      for (int j = 0; j < D; ++j) {
         Y[i][j] = Yn[i][j];
         Z[i][j] = Zn[i][j];
       }
-    if (delta < 1e-32) break;  // converged (T ~ I)
+    if (delta < 1e-30) break;  // converged (T ~ I)
   }
   const double c = rsqrt(tr);
-  for (int i = 0; i < d; ++i)
-    for (int j = 0; j < d; ++j) out[i][j] = Z[i][j] * c;
+  #pragma unroll
+  for (int i = 0; i < D; ++i)
+    #pragma unroll
+    for (int j = 0; j < D; ++j) out[i][j] = Z[i][j] * c;
 }
 
+template <int D, int R>
 __global__ void k_polar_affine(const double* __restrict__ A,
                                const double* __restrict__ B,
                                const double* __restrict__ C,
                                double ca, double cb, double cc,
                                double* __restrict__ out,
-                               int n, int d, int r,
-                               const double* __restrict__ ctrl, int guard) {
+                               int n, const double* __restrict__ ctrl,
+                               int guard) {
   if (guarded_off(ctrl, guard)) return;
   const int i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
-  const int dh = d + 1;
-  double Mt[DPO_MAX_DH][DPO_MAX_R];
-  const double* Ai = A + (size_t)i * dh * r;
-  const double* Bi = B ? B + (size_t)i * dh * r : nullptr;
-  const double* Ci = C ? C + (size_t)i * dh * r : nullptr;
+  constexpr int dh = D + 1;
+  double Mt[dh][R];
+  const double* Ai = A + (size_t)i * dh * R;
+  const double* Bi = B ? B + (size_t)i * dh * R : nullptr;
+  const double* Ci = C ? C + (size_t)i * dh * R : nullptr;
+  #pragma unroll
   for (int c = 0; c < dh; ++c)
-    for (int k = 0; k < r; ++k) {
-      double v = ca * Ai[c * r + k];
-      if (Bi) v = fma(cb, Bi[c * r + k], v);
-      if (Ci) v = fma(cc, Ci[c * r + k], v);
+    #pragma unroll
+    for (int k = 0; k < R; ++k) {
+      double v = ca * Ai[c * R + k];
+      if (Bi) v = fma(cb, Bi[c * R + k], v);
+      if (Ci) v = fma(cc, Ci[c * R + k], v);
       Mt[c][k] = v;
     }
-  // Gram = Mt Mt^T (d x d); polar(Mt) = Gram^{-1/2} Mt
-  double S[3][3] = {};
-  for (int a = 0; a < d; ++a)
-    for (int b = 0; b < d; ++b) {
+  // Gram = Mt Mt^T (D x D); polar(Mt) = Gram^{-1/2} Mt
+  double S[D][D];
+  #pragma unroll
+  for (int a = 0; a < D; ++a)
+    #pragma unroll
+    for (int b = 0; b < D; ++b) {
       double s = 0.0;
-      for (int k = 0; k < r; ++k) s = fma(Mt[a][k], Mt[b][k], s);
+      #pragma unroll
+      for (int k = 0; k < R; ++k) s = fma(Mt[a][k], Mt[b][k], s);
       S[a][b] = s;
     }
-  double Gi[3][3];
-  spd_inv_sqrt(S, d, Gi);
-  double* Oi = out + (size_t)i * dh * r;
-  for (int a = 0; a < d; ++a)
-    for (int k = 0; k < r; ++k) {
+  double Gi[D][D];
+  spd_inv_sqrt<D>(S, Gi);
+  double* Oi = out + (size_t)i * dh * R;
+  #pragma unroll
+  for (int a = 0; a < D; ++a)
+    #pragma unroll
+    for (int k = 0; k < R; ++k) {
       double s = 0.0;
-      for (int b = 0; b < d; ++b) s = fma(Gi[a][b], Mt[b][k], s);
-      Oi[a * r + k] = s;
+      #pragma unroll
+      for (int b = 0; b < D; ++b) s = fma(Gi[a][b], Mt[b][k], s);
+      Oi[a * R + k] = s;
     }
-  for (int k = 0; k < r; ++k) Oi[d * r + k] = Mt[d][k];
+  #pragma unroll
+  for (int k = 0; k < R; ++k) Oi[D * R + k] = Mt[D][k];
 }
 
 // ---------------------------------------------------------------------
@@ -336,61 +406,88 @@ __global__ void k_polar_affine(const double* __restrict__ A,
 // threads (i) are coalesced; j-loop broadcasts V[j][*] via LDS.
 // fp32 storage halves the bandwidth; accumulate fp64.
 // ---------------------------------------------------------------------
+// Z = Minv @ V with Minv symmetric fp32 read column-wise (coalesced).
+// The j loop is SPLIT across gridDim.y so small problems still fill the
+// chip (MI355X: 256 CUs want >> 256 workgroups); partial sums combine
+// with fp64 atomics into the zeroed output.
+template <int R>
 __global__ void k_precond_dense(const float* __restrict__ Minv,
                                 const double* __restrict__ V,
                                 double* __restrict__ Z,
-                                int N, int r,
-                                const double* __restrict__ ctrl, int guard) {
+                                int N, const double* __restrict__ ctrl,
+                                int guard) {
   if (guarded_off(ctrl, guard)) return;
   const int i = blockIdx.x * blockDim.x + threadIdx.x;
-  extern __shared__ double shv[];  // TILE_J * r
-  const int TILE_J = 64;
-  double acc[DPO_MAX_R] = {};
-  for (int j0 = 0; j0 < N; j0 += TILE_J) {
-    const int jn = min(TILE_J, N - j0);
+  const int jsplit = gridDim.y;
+  const int jchunk = (N + jsplit - 1) / jsplit;
+  const int j_lo = blockIdx.y * jchunk;
+  const int j_hi = min(N, j_lo + jchunk);
+  constexpr int TILE_J = 64;
+  __shared__ double shv[TILE_J * R];
+  double acc[R];
+  #pragma unroll
+  for (int k = 0; k < R; ++k) acc[k] = 0.0;
+  for (int j0 = j_lo; j0 < j_hi; j0 += TILE_J) {
+    const int jn = min(TILE_J, j_hi - j0);
     __syncthreads();
-    for (int t = threadIdx.x; t < jn * r; t += blockDim.x)
-      shv[t] = V[(size_t)(j0 + t / r) * r + (t % r)];
+    for (int t = threadIdx.x; t < jn * R; t += blockDim.x)
+      shv[t] = V[(size_t)(j0 + t / R) * R + (t % R)];
     __syncthreads();
     if (i < N) {
       for (int j = 0; j < jn; ++j) {
         const double m = (double)Minv[(size_t)(j0 + j) * N + i];
         #pragma unroll
-        for (int k = 0; k < DPO_MAX_R; ++k)
-          if (k < r) acc[k] = fma(m, shv[j * r + k], acc[k]);
+        for (int k = 0; k < R; ++k)
+          acc[k] = fma(m, shv[j * R + k], acc[k]);
       }
     }
   }
-  if (i < N)
-    for (int k = 0; k < r; ++k) Z[(size_t)i * r + k] = acc[k];
+  if (i < N) {
+    if (jsplit == 1) {
+      #pragma unroll
+      for (int k = 0; k < R; ++k) Z[(size_t)i * R + k] = acc[k];
+    } else {
+      #pragma unroll
+      for (int k = 0; k < R; ++k)
+        atomicAdd(&Z[(size_t)i * R + k], acc[k]);
+    }
+  }
 }
 
 // Block-Jacobi apply: per pose solve (L L^T) z = v with stored Cholesky
 // factors L (n, dh, dh). One thread per pose per rhs column.
+template <int DH>
 __global__ void k_precond_jacobi(const double* __restrict__ L,
                                  const double* __restrict__ V,
                                  double* __restrict__ Z,
-                                 int n, int dh, int r,
+                                 int n, int r,
                                  const double* __restrict__ ctrl, int guard) {
   if (guarded_off(ctrl, guard)) return;
   const int idx = blockIdx.x * blockDim.x + threadIdx.x;
   if (idx >= n * r) return;
   const int i = idx / r, k = idx % r;
-  const double* Li = L + (size_t)i * dh * dh;
-  double y[DPO_MAX_DH];
+  const double* Li = L + (size_t)i * DH * DH;
+  double y[DH];
   // forward solve L y = v
-  for (int a = 0; a < dh; ++a) {
-    double s = V[((size_t)i * dh + a) * r + k];
-    for (int b = 0; b < a; ++b) s = fma(-Li[a * dh + b], y[b], s);
-    y[a] = s / Li[a * dh + a];
+  #pragma unroll
+  for (int a = 0; a < DH; ++a) {
+    double s = V[((size_t)i * DH + a) * r + k];
+    #pragma unroll
+    for (int b = 0; b < DH; ++b)
+      if (b < a) s = fma(-Li[a * DH + b], y[b], s);
+    y[a] = s / Li[a * DH + a];
   }
   // backward solve L^T z = y
-  for (int a = dh - 1; a >= 0; --a) {
+  #pragma unroll
+  for (int a = DH - 1; a >= 0; --a) {
     double s = y[a];
-    for (int b = a + 1; b < dh; ++b) s = fma(-Li[b * dh + a], y[b], s);
-    y[a] = s / Li[a * dh + a];
+    #pragma unroll
+    for (int b = 0; b < DH; ++b)
+      if (b > a) s = fma(-Li[b * DH + a], y[b], s);
+    y[a] = s / Li[a * DH + a];
   }
-  for (int a = 0; a < dh; ++a) Z[((size_t)i * dh + a) * r + k] = y[a];
+  #pragma unroll
+  for (int a = 0; a < DH; ++a) Z[((size_t)i * DH + a) * r + k] = y[a];
 }
 
 // ---------------------------------------------------------------------
@@ -710,10 +807,109 @@ __global__ void k_g_assemble(double* __restrict__ Gt,
 }
 
 // ---------------------------------------------------------------------
-// C ABI
+// (d, r) dispatch: the supported compile-time shapes. SE(2): d=2,
+// r in 2..6; SE(3): d=3, r in 3..8. Everything the reference exercises
+// (r=5 RBCD, r=d batch) is covered; exotic shapes abort loudly.
 // ---------------------------------------------------------------------
+#define DPO_FOREACH_DR(F) \
+  F(2, 2) F(2, 3) F(2, 4) F(2, 5) F(2, 6) \
+  F(3, 3) F(3, 4) F(3, 5) F(3, 6) F(3, 7) F(3, 8)
+
 static inline int blocks_for(long total, int bs) {
   return (int)((total + bs - 1) / bs);
+}
+
+static void dpo_bad_shape(int d, int r) {
+  fprintf(stderr, "dpo_ops: unsupported (d=%d, r=%d) kernel shape\n", d, r);
+  abort();
+}
+
+static void launch_spmm(const int* rp, const int* ci, const double* vals,
+                        int n, int d, int r, const double* X, double* out,
+                        const double* ctrl, int guard, hipStream_t s) {
+  const int dh = d + 1;
+  const int tile = dh * r;
+  const int grid = blocks_for(n, 256 / tile);
+#define CASE_SPMM(D, R) \
+  if (d == D && r == R) { \
+    hipLaunchKernelGGL((k_bsr_spmm<D + 1, R>), dim3(grid), dim3(256), 0, s, \
+                       rp, ci, vals, X, out, n, ctrl, guard); \
+    return; \
+  }
+  DPO_FOREACH_DR(CASE_SPMM)
+#undef CASE_SPMM
+  hipLaunchKernelGGL(k_bsr_spmm_gen, dim3(grid), dim3(256), 0, s,
+                     rp, ci, vals, X, out, n, dh, r, ctrl, guard);
+}
+
+static void launch_proj_dots(const double* X, const double* V,
+                             const double* G, double* out,
+                             const double* dotWith, double* ctrl,
+                             int n, int d, int r, int dot_slot,
+                             int dot_slot2, int guard, hipStream_t s) {
+  const int grid = blocks_for(n, 256);
+#define CASE_PROJ(D, R) \
+  if (d == D && r == R) { \
+    hipLaunchKernelGGL((k_proj_dots<0, D, R>), dim3(grid), dim3(256), 0, s, \
+                       X, V, G, out, dotWith, ctrl, n, dot_slot, \
+                       dot_slot2, guard); \
+    return; \
+  }
+  DPO_FOREACH_DR(CASE_PROJ)
+#undef CASE_PROJ
+  dpo_bad_shape(d, r);
+}
+
+static void launch_polar(const double* A, const double* B, const double* C,
+                         double ca, double cb, double cc, double* out,
+                         int n, int d, int r, const double* ctrl, int guard,
+                         hipStream_t s) {
+  const int grid = blocks_for(n, 256);
+#define CASE_POLAR(D, R) \
+  if (d == D && r == R) { \
+    hipLaunchKernelGGL((k_polar_affine<D, R>), dim3(grid), dim3(256), 0, s, \
+                       A, B, C, ca, cb, cc, out, n, ctrl, guard); \
+    return; \
+  }
+  DPO_FOREACH_DR(CASE_POLAR)
+#undef CASE_POLAR
+  dpo_bad_shape(d, r);
+}
+
+static void launch_precond_dense(const float* Minv, const double* V,
+                                 double* Z, int N, int r,
+                                 const double* ctrl, int guard,
+                                 hipStream_t s) {
+  const int gx = blocks_for(N, 256);
+  // j-split so small problems still produce >= ~512 workgroups
+  int jsplit = 1;
+  while (gx * jsplit < 512 && jsplit < 32) jsplit *= 2;
+  if (jsplit > 1)
+    DPO_CHECK(hipMemsetAsync(Z, 0, (size_t)N * r * sizeof(double), s));
+#define CASE_PD(D, R) \
+  if (r == R) { \
+    hipLaunchKernelGGL((k_precond_dense<R>), dim3(gx, jsplit), dim3(256), \
+                       0, s, Minv, V, Z, N, ctrl, guard); \
+    return; \
+  }
+  DPO_FOREACH_DR(CASE_PD)
+#undef CASE_PD
+  dpo_bad_shape(-1, r);
+}
+
+static void launch_precond_jacobi(const double* L, const double* V,
+                                  double* Z, int n, int dh, int r,
+                                  const double* ctrl, int guard,
+                                  hipStream_t s) {
+  const int grid = blocks_for((long)n * r, 256);
+  if (dh == 3)
+    hipLaunchKernelGGL((k_precond_jacobi<3>), dim3(grid), dim3(256), 0, s,
+                       L, V, Z, n, r, ctrl, guard);
+  else if (dh == 4)
+    hipLaunchKernelGGL((k_precond_jacobi<4>), dim3(grid), dim3(256), 0, s,
+                       L, V, Z, n, r, ctrl, guard);
+  else
+    dpo_bad_shape(dh - 1, r);
 }
 
 extern "C" {
@@ -721,57 +917,38 @@ extern "C" {
 void dpo_bsr_spmm(const int* row_ptr, const int* col_idx, const double* vals,
                   int n, int dh, const double* X, double* out, int r,
                   const double* ctrl, int guard, void* stream) {
-  hipStream_t s = (hipStream_t)stream;
-  const int tile = dh * r;
-  const int per_block = 256 / tile;
-  const int grid = blocks_for(n, per_block);
-  hipLaunchKernelGGL(k_bsr_spmm, dim3(grid), dim3(256), 0, s,
-                     row_ptr, col_idx, vals, X, out, n, dh, r, ctrl, guard);
+  launch_spmm(row_ptr, col_idx, vals, n, dh - 1, r, X, out, ctrl, guard,
+              (hipStream_t)stream);
 }
 
 void dpo_proj_dots(const double* X, const double* V, const double* G,
                    double* out, const double* dotWith, double* ctrl,
                    int n, int d, int r, int dot_slot, int dot_slot2,
                    int neg, int guard, void* stream) {
-  hipStream_t s = (hipStream_t)stream;
-  const int grid = blocks_for(n, 256);
-  if (neg)
-    hipLaunchKernelGGL(k_proj_dots<1>, dim3(grid), dim3(256), 0, s,
-                       X, V, G, out, dotWith, ctrl, n, d, r,
-                       dot_slot, dot_slot2, guard);
-  else
-    hipLaunchKernelGGL(k_proj_dots<0>, dim3(grid), dim3(256), 0, s,
-                       X, V, G, out, dotWith, ctrl, n, d, r,
-                       dot_slot, dot_slot2, guard);
+  (void)neg;  // the negated variant is realized by k_tcg_delta instead
+  launch_proj_dots(X, V, G, out, dotWith, ctrl, n, d, r, dot_slot,
+                   dot_slot2, guard, (hipStream_t)stream);
 }
 
 void dpo_polar_affine(const double* A, const double* B, const double* C,
                       double ca, double cb, double cc, double* out,
                       int n, int d, int r, const double* ctrl, int guard,
                       void* stream) {
-  hipStream_t s = (hipStream_t)stream;
-  const int grid = blocks_for(n, 256);
-  hipLaunchKernelGGL(k_polar_affine, dim3(grid), dim3(256), 0, s,
-                     A, B, C, ca, cb, cc, out, n, d, r, ctrl, guard);
+  launch_polar(A, B, C, ca, cb, cc, out, n, d, r, ctrl, guard,
+               (hipStream_t)stream);
 }
 
 void dpo_precond_dense(const float* Minv, const double* V, double* Z,
                        int N, int r, const double* ctrl, int guard,
                        void* stream) {
-  hipStream_t s = (hipStream_t)stream;
-  const int grid = blocks_for(N, 256);
-  const size_t shmem = 64 * r * sizeof(double);
-  hipLaunchKernelGGL(k_precond_dense, dim3(grid), dim3(256), shmem, s,
-                     Minv, V, Z, N, r, ctrl, guard);
+  launch_precond_dense(Minv, V, Z, N, r, ctrl, guard, (hipStream_t)stream);
 }
 
 void dpo_precond_jacobi(const double* L, const double* V, double* Z,
                         int n, int dh, int r, const double* ctrl, int guard,
                         void* stream) {
-  hipStream_t s = (hipStream_t)stream;
-  const int grid = blocks_for((long)n * r, 256);
-  hipLaunchKernelGGL(k_precond_jacobi, dim3(grid), dim3(256), 0, s,
-                     L, V, Z, n, dh, r, ctrl, guard);
+  launch_precond_jacobi(L, V, Z, n, dh, r, ctrl, guard,
+                        (hipStream_t)stream);
 }
 
 void dpo_tcg_update(double* eta, double* rvec, const double* delta,
@@ -925,26 +1102,17 @@ static void ctx_assemble_g(DpoCtx* c, const double* nbr, hipStream_t s) {
 
 static void ctx_precond(DpoCtx* c, const double* V, double* Z,
                         hipStream_t s) {
-  if (c->Minv) {
-    const int grid = (c->N + 255) / 256;
-    const size_t shmem = 64 * c->r * sizeof(double);
-    hipLaunchKernelGGL(k_precond_dense, dim3(grid), dim3(256), shmem, s,
-                       c->Minv, V, Z, c->N, c->r, c->ctrl, ST_RUN);
-  } else {
-    const int grid = ((long)c->n * c->r + 255) / 256;
-    hipLaunchKernelGGL(k_precond_jacobi, dim3(grid), dim3(256), 0, s,
-                       c->Ljac, V, Z, c->n, c->dh, c->r, c->ctrl, ST_RUN);
-  }
+  if (c->Minv)
+    launch_precond_dense(c->Minv, V, Z, c->N, c->r, c->ctrl, ST_RUN, s);
+  else
+    launch_precond_jacobi(c->Ljac, V, Z, c->n, c->dh, c->r, c->ctrl,
+                          ST_RUN, s);
 }
 
 static void ctx_spmm(DpoCtx* c, const double* X, double* out, int guard,
                      hipStream_t s) {
-  const int tile = c->dh * c->r;
-  const int per_block = 256 / tile;
-  const int grid = (c->n + per_block - 1) / per_block;
-  hipLaunchKernelGGL(k_bsr_spmm, dim3(grid), dim3(256), 0, s,
-                     c->q_rp, c->q_ci, c->q_vals, X, out, c->n, c->dh,
-                     c->r, c->ctrl, guard);
+  launch_spmm(c->q_rp, c->q_ci, c->q_vals, c->n, c->d, c->r, X, out,
+              c->ctrl, guard, s);
 }
 
 extern "C" {
@@ -1006,9 +1174,8 @@ int dpo_rbcd_solve(void* h, double* X, double tol, double Delta0,
 
   // gradient phase
   ctx_spmm(c, X, c->W, -1, s);
-  hipLaunchKernelGGL(k_proj_dots<0>, dim3((n + 255) / 256), dim3(256), 0, s,
-                     X, c->W, c->Gt, c->grad, (const double*)nullptr,
-                     c->ctrl, n, d, r, C_DOT1, C_DOT0, -1);
+  launch_proj_dots(X, c->W, c->Gt, c->grad, nullptr, c->ctrl, n,
+                   d, r, C_DOT1, C_DOT0, -1, s);
   if (c->Gt)
     hipLaunchKernelGGL(k_dots, dim3(gvec), dim3(256), 0, s,
                        c->Gt, X, (const double*)nullptr, c->ctrl,
@@ -1020,9 +1187,8 @@ int dpo_rbcd_solve(void* h, double* X, double tol, double Delta0,
                      Delta0, 1.0, 0.1);
   // z0
   ctx_precond(c, c->rvec, c->z, s);
-  hipLaunchKernelGGL(k_proj_dots<0>, dim3((n + 255) / 256), dim3(256), 0, s,
-                     X, c->z, (const double*)nullptr, c->z, c->rvec,
-                     c->ctrl, n, d, r, C_DOT0, -1, ST_RUN);
+  launch_proj_dots(X, c->z, nullptr, c->z, c->rvec, c->ctrl, n, d, r,
+                   C_DOT0, -1, ST_RUN, s);
   hipLaunchKernelGGL(k_ctrl_z0, dim3(1), dim3(64), 0, s, c->ctrl);
   hipLaunchKernelGGL(k_tcg_delta, dim3(gvec), dim3(256), 0, s,
                      c->delta, c->z, c->ctrl, total);
@@ -1030,18 +1196,16 @@ int dpo_rbcd_solve(void* h, double* X, double tol, double Delta0,
   // tCG loop
   for (int j = 0; j < c->max_inner; ++j) {
     ctx_spmm(c, c->delta, c->Hd, ST_RUN, s);
-    hipLaunchKernelGGL(k_proj_dots<0>, dim3((n + 255) / 256), dim3(256), 0,
-                       s, X, c->Hd, (const double*)nullptr, c->Hd,
-                       c->delta, c->ctrl, n, d, r, C_DOT0, -1, ST_RUN);
+    launch_proj_dots(X, c->Hd, nullptr, c->Hd, c->delta, c->ctrl, n, d,
+                     r, C_DOT0, -1, ST_RUN, s);
     hipLaunchKernelGGL(k_ctrl_alpha, dim3(1), dim3(64), 0, s, c->ctrl);
     hipLaunchKernelGGL(k_tcg_update, dim3(gvec), dim3(256), 0, s,
                        c->eta, c->rvec, c->delta, c->Hd, c->eta_snap,
                        c->delta_snap, c->ctrl, total);
     hipLaunchKernelGGL(k_ctrl_rr, dim3(1), dim3(64), 0, s, c->ctrl);
     ctx_precond(c, c->rvec, c->z, s);
-    hipLaunchKernelGGL(k_proj_dots<0>, dim3((n + 255) / 256), dim3(256), 0,
-                       s, X, c->z, (const double*)nullptr, c->z, c->rvec,
-                       c->ctrl, n, d, r, C_DOT0, -1, ST_RUN);
+    launch_proj_dots(X, c->z, nullptr, c->z, c->rvec, c->ctrl, n, d, r,
+                     C_DOT0, -1, ST_RUN, s);
     hipLaunchKernelGGL(k_ctrl_beta, dim3(1), dim3(64), 0, s, c->ctrl);
     hipLaunchKernelGGL(k_tcg_delta, dim3(gvec), dim3(256), 0, s,
                        c->delta, c->z, c->ctrl, total);
@@ -1057,9 +1221,8 @@ int dpo_rbcd_solve(void* h, double* X, double tol, double Delta0,
     hipLaunchKernelGGL(k_form_step, dim3(gvec), dim3(256), 0, s,
                        c->step, c->eta, c->eta_snap, c->delta_snap,
                        c->ctrl, total);
-    hipLaunchKernelGGL(k_polar_affine, dim3((n + 255) / 256), dim3(256), 0,
-                       s, X, c->step, (const double*)nullptr, 1.0, 1.0, 0.0,
-                       c->Xprop, n, d, r, c->ctrl, ST_TCG_STOP);
+    launch_polar(X, c->step, nullptr, 1.0, 1.0, 0.0, c->Xprop, n, d, r,
+                 c->ctrl, ST_TCG_STOP, s);
     ctx_spmm(c, c->Xprop, c->W, ST_TCG_STOP, s);
     hipLaunchKernelGGL(k_dots, dim3(gvec), dim3(256), 0, s,
                        c->Xprop, c->W, c->Gt, c->ctrl, C_DOT0, C_DOT2,
@@ -1089,9 +1252,8 @@ int dpo_rbcd_solve(void* h, double* X, double tol, double Delta0,
   if (compute_final_gn && status == ST_ACCEPTED) {
     DPO_CHECK(hipMemsetAsync(c->ctrl + C_DOT1, 0, sizeof(double), s));
     ctx_spmm(c, X, c->W, -1, s);
-    hipLaunchKernelGGL(k_proj_dots<0>, dim3((n + 255) / 256), dim3(256), 0,
-                       s, X, c->W, c->Gt, c->grad, (const double*)nullptr,
-                       c->ctrl, n, d, r, C_DOT1, -1, -1);
+    launch_proj_dots(X, c->W, c->Gt, c->grad, nullptr, c->ctrl, n, d, r,
+                     C_DOT1, -1, -1, s);
     DPO_CHECK(hipMemcpyAsync(c->ctrl_host + C_DOT1, c->ctrl + C_DOT1,
                              sizeof(double), hipMemcpyDeviceToHost, s));
     DPO_CHECK(hipStreamSynchronize(s));
@@ -1121,9 +1283,8 @@ void dpo_eval_terms(void* h, const double* X, double* out_dev,
   const int gvec = (int)((total + 255) / 256);
   DPO_CHECK(hipMemsetAsync(c->ctrl + C_DOT0, 0, 4 * sizeof(double), s));
   ctx_spmm(c, X, c->W, -1, s);
-  hipLaunchKernelGGL(k_proj_dots<0>, dim3((n + 255) / 256), dim3(256), 0, s,
-                     X, c->W, c->Gt, c->grad, (const double*)nullptr,
-                     c->ctrl, n, d, r, C_DOT1, C_DOT0, -1);
+  launch_proj_dots(X, c->W, c->Gt, c->grad, nullptr, c->ctrl, n,
+                   d, r, C_DOT1, C_DOT0, -1, s);
   if (c->Gt)
     hipLaunchKernelGGL(k_dots, dim3(gvec), dim3(256), 0, s,
                        c->Gt, X, (const double*)nullptr, c->ctrl,
