@@ -11,7 +11,7 @@ import subprocess
 import sys
 
 HIP_DIR = os.path.dirname(os.path.abspath(__file__)) + "/hip"
-SOURCES = ["dpo_ops.hip"]
+SOURCES = ["dpo_ops.hip", "dpo_partition.cpp"]
 OUT = os.path.join(HIP_DIR, "libdpo_hip_ops.so")
 ARCH = os.environ.get("DPO_GFX_ARCH", "gfx950")
 

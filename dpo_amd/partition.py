@@ -18,6 +18,34 @@ import numpy as np
 from .types import RelativeSEMeasurement
 
 
+def _native_multilevel(adj_lists, k, imbalance, seed, n_restarts):
+    try:
+        import ctypes
+        import os
+        lib_path = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                                "ops", "hip", "libdpo_hip_ops.so")
+        lib = ctypes.CDLL(lib_path)
+        fn = lib.dpo_partition_multilevel
+        fn.restype = ctypes.c_double
+        fn.argtypes = [ctypes.c_int, ctypes.c_void_p, ctypes.c_void_p,
+                       ctypes.c_void_p, ctypes.c_int, ctypes.c_double,
+                       ctypes.c_int, ctypes.c_uint, ctypes.c_void_p]
+    except OSError:
+        return None
+    n = len(adj_lists)
+    xadj = np.zeros(n + 1, dtype=np.int32)
+    for u, a in enumerate(adj_lists):
+        xadj[u + 1] = xadj[u] + len(a)
+    adjncy = np.zeros(max(int(xadj[n]), 1), dtype=np.int32)
+    for u, a in enumerate(adj_lists):
+        adjncy[xadj[u]:xadj[u + 1]] = a
+    out = np.zeros(n, dtype=np.int32)
+    fn(n, xadj.ctypes.data, adjncy.ctypes.data, None, k,
+       float(imbalance), int(n_restarts), int(seed) & 0xFFFFFFFF,
+       out.ctypes.data)
+    return [int(x) for x in out]
+
+
 def contiguous_partition(num_poses: int, k: int) -> List[int]:
     """Naive sequential chunking (reference MultiRobotExample.cpp:93-110)."""
     per = num_poses // k
@@ -219,10 +247,19 @@ def _fm_refine(g: _Graph, part: np.ndarray, k: int, max_wgt: float,
 def multilevel_partition(adj_lists: Sequence[Sequence[int]], k: int,
                          imbalance: float = 0.05,
                          coarsen_to: int = 0,
-                         seed: int = 1) -> List[int]:
-    """Multi-level k-way partition of an undirected graph."""
+                         seed: int = 1,
+                         n_restarts: int = 8) -> List[int]:
+    """Multi-level k-way partition of an undirected graph.
+
+    Uses the native C++ partitioner (heavy-edge matching + graph growing
+    + hill-climbing FM with rollback + connectivity fixup,
+    dpo_partition.cpp) when the extension is built; falls back to the
+    pure-Python implementation below."""
     if k <= 1:
         return [0] * len(adj_lists)
+    native = _native_multilevel(adj_lists, k, imbalance, seed, n_restarts)
+    if native is not None:
+        return native
     g0 = _build_graph(adj_lists)
     total = float(g0.vwgt.sum())
     max_wgt = (1.0 + imbalance) * total / k
