@@ -208,10 +208,6 @@ class PGOAgent:
             [0 if m.r1 == self.id else 1 for m in self.shared_lc],
             nbr_slots)
 
-    def _weights_tensor(self) -> Tensor:
-        return torch.tensor([m.weight for m in self._all_meas],
-                            dtype=torch.float64)
-
     def _construct_q(self) -> None:
         Q = self._q_assembler.assemble(self._weights_tensor())
         if self.device.type != "cpu":
@@ -766,6 +762,87 @@ class PGOAgent:
         self.X = None
 
     # ------------------------------------------------------------------
+    # SoA setup for large graphs (MeasurementArray end-to-end; no
+    # per-edge Python objects). Feeds the packed GPU path incl. robust.
+    # ------------------------------------------------------------------
+    def set_pose_graph_arrays(self, odometry_ma, private_ma, shared_ma,
+                              T_init: Optional[np.ndarray] = None) -> None:
+        """Array-based set_pose_graph for large graphs. odometry/private/
+        shared are MeasurementArrays in LOCAL indices (r1/r2 = robot ids).
+        Robust (non-L2) agents use vectorized odometry initialization."""
+        from .measurements import (MeasurementArray,
+                                   concat_measurement_arrays,
+                                   odometry_initialization_array)
+        assert self.state == PGOAgentState.WAIT_FOR_DATA
+        self._soa = True
+        d = self.d
+        odometry_ma.is_known_inlier[:] = True
+        private_ma.is_known_inlier[:] = False
+        shared_ma.is_known_inlier[:] = False
+        self._odo_ma, self._priv_ma, self._shared_ma = \
+            odometry_ma, private_ma, shared_ma
+        n = 1
+        for ma in (odometry_ma, private_ma):
+            if len(ma):
+                n = max(n, int(ma.p1.max()) + 1, int(ma.p2.max()) + 1)
+        sh = shared_ma
+        if len(sh):
+            mine1 = sh.r1 == self.id
+            local_p = np.where(mine1, sh.p1, sh.p2)
+            nbr_r = np.where(mine1, sh.r2, sh.r1)
+            nbr_p = np.where(mine1, sh.p2, sh.p1)
+            n = max(n, int(local_p.max()) + 1)
+            self.local_shared_pose_ids = {
+                (self.id, int(p)) for p in np.unique(local_p)}
+            nbr_ids = np.unique(np.stack([nbr_r, nbr_p], 1), axis=0)
+            self.neighbor_shared_pose_ids = {
+                (int(a), int(b)) for a, b in nbr_ids}
+            self.neighbor_robot_ids = {int(x) for x in np.unique(nbr_r)}
+        self.n = n
+        self._manifold = LiftedSEManifold(self.r, d, n)
+        self.problem = QuadraticProblem(n, d, self.r)
+
+        npriv = len(odometry_ma) + len(private_ma)
+        nsh = len(shared_ma)
+        all_ma = concat_measurement_arrays(
+            [odometry_ma, private_ma, shared_ma])
+        shared_flags = [False] * npriv + [True] * nsh
+        lep = [0] * npriv + [0 if r1 == self.id else 1 for r1 in sh.r1]
+        self._all_ma = all_ma
+        self._q_assembler = QAssembler(n, d, all_ma, shared_flags, lep)
+        self._nbr_slot_order = sorted(self.neighbor_shared_pose_ids)
+        slot_of = {pid: k for k, pid in enumerate(self._nbr_slot_order)}
+        if nsh:
+            nbr_slots = [slot_of[(int(a), int(b))]
+                         for a, b in zip(nbr_r, nbr_p)]
+        else:
+            nbr_slots = []
+        self._g_assembler = GAssembler(
+            n, d, shared_ma,
+            [0 if r1 == self.id else 1 for r1 in sh.r1], nbr_slots)
+        self._construct_q()
+        if T_init is not None:
+            self.T_local_init = T_init
+        elif self.params.robust_cost_type != RobustCostType.L2:
+            self.T_local_init = odometry_initialization_array(
+                d, n, odometry_ma)
+        else:
+            self.T_local_init = chordal_initialization(
+                d, n, concat_measurement_arrays(
+                    [odometry_ma, private_ma]).to_list())
+        assert self.YLift is not None
+        Xt = _T_to_Xt(self.T_local_init, self.r, self.YLift)
+        self.X = torch.from_numpy(Xt).to(self.device)
+        self.XInit = self.X.clone()
+        self.state = PGOAgentState.INITIALIZED
+
+    def _weights_tensor(self) -> Tensor:
+        if getattr(self, "_soa", False):
+            return torch.from_numpy(self._all_ma.weight.copy())
+        return torch.tensor([m.weight for m in self._all_meas],
+                            dtype=torch.float64)
+
+    # ------------------------------------------------------------------
     # packed fast path (GPU): device-tensor neighbor buffers + native
     # C++ solve/eval, driven by DistributedRBCDDriver._run_packed.
     # ------------------------------------------------------------------
@@ -776,9 +853,16 @@ class PGOAgent:
         self._nbr_buffer = torch.zeros(max(n_slots, 1), self.dh, self.r,
                                        dtype=torch.float64, device=dev)
         self._nbr_buffer_aux = torch.zeros_like(self._nbr_buffer)
-        self._w_shared_dev = torch.tensor(
-            [m.weight for m in self.shared_lc], dtype=torch.float64,
-            device=dev)
+        # full per-measurement weight vector on device (odometry |
+        # private | shared); the shared tail doubles as the G-assembly
+        # weight view and the exchange payload.
+        self._all_weights_dev = self._weights_tensor().to(dev)
+        if getattr(self, "_soa", False):
+            nsh = len(self._shared_ma)
+        else:
+            nsh = len(self.shared_lc)
+        self._shared_w_off = self._all_weights_dev.numel() - nsh
+        self._w_shared_dev = self._all_weights_dev[self._shared_w_off:]
         if getattr(self, "_dev_solver", None) is None:
             from .ops.hip_backend import DeviceSolver
             self._dev_solver = DeviceSolver(self.n, self.d, self.r, dev,
@@ -814,6 +898,107 @@ class PGOAgent:
     def _packed_eval(self):
         """Device 3-vector [f, 0.5<X,G>, ||rgrad||^2] with fresh G."""
         return self._dev_solver.round_eval(self.X, self._nbr_buffer)
+
+    # --- packed robust (GNC) support ----------------------------------
+    def _packed_gnc_setup(self, dev) -> None:
+        """Device tensors for the GNC weight-update kernel. SoA mode."""
+        if getattr(self, "_gnc_ready", False):
+            return
+        from .measurements import concat_measurement_arrays
+        assert getattr(self, "_soa", False), \
+            "packed robust mode requires set_pose_graph_arrays"
+        priv, sh = self._priv_ma, self._shared_ma
+        n_odo = len(self._odo_ma)
+        n_priv = len(priv)
+        slot_of = {pid: k for k, pid in enumerate(self._nbr_slot_order)}
+        import numpy as np
+        ne = n_priv + len(sh)
+        e1_idx = np.zeros(ne, dtype=np.int64)
+        e2_idx = np.zeros(ne, dtype=np.int64)
+        e1_nbr = np.zeros(ne, dtype=np.uint8)
+        e2_nbr = np.zeros(ne, dtype=np.uint8)
+        upd = np.zeros(ne, dtype=np.uint8)
+        widx = np.zeros(ne, dtype=np.int64)
+        R = np.zeros((ne, self.d, self.d))
+        t = np.zeros((ne, self.d))
+        kap = np.zeros(ne)
+        tau = np.zeros(ne)
+        if n_priv:
+            e1_idx[:n_priv] = priv.p1
+            e2_idx[:n_priv] = priv.p2
+            upd[:n_priv] = (~priv.is_known_inlier).astype(np.uint8)
+            widx[:n_priv] = n_odo + np.arange(n_priv)
+            R[:n_priv] = priv.R
+            t[:n_priv] = priv.t
+            kap[:n_priv] = priv.kappa
+            tau[:n_priv] = priv.tau
+        if len(sh):
+            mine1 = sh.r1 == self.id
+            other = np.where(mine1, sh.r2, sh.r1)
+            for k in range(len(sh)):
+                e = n_priv + k
+                if mine1[k]:
+                    e1_idx[e] = sh.p1[k]
+                    e2_idx[e] = slot_of[(int(sh.r2[k]), int(sh.p2[k]))]
+                    e2_nbr[e] = 1
+                else:
+                    e1_idx[e] = slot_of[(int(sh.r1[k]), int(sh.p1[k]))]
+                    e1_nbr[e] = 1
+                    e2_idx[e] = sh.p2[k]
+            # owner-computes: this agent updates edges whose OTHER robot
+            # has a larger id (PGOAgent.cpp:1201-1235)
+            upd[n_priv:] = ((other > self.id)
+                            & ~sh.is_known_inlier).astype(np.uint8)
+            widx[n_priv:] = n_odo + n_priv + np.arange(len(sh))
+            R[n_priv:] = sh.R
+            t[n_priv:] = sh.t
+            kap[n_priv:] = sh.kappa
+            tau[n_priv:] = sh.tau
+        self._gnc = {
+            "e1_idx": torch.from_numpy(e1_idx).to(dev),
+            "e2_idx": torch.from_numpy(e2_idx).to(dev),
+            "e1_nbr": torch.from_numpy(e1_nbr).to(dev),
+            "e2_nbr": torch.from_numpy(e2_nbr).to(dev),
+            "upd": torch.from_numpy(upd).to(dev),
+            "widx": torch.from_numpy(widx).to(dev),
+            "R": torch.from_numpy(R).to(dev).contiguous(),
+            "t": torch.from_numpy(t).to(dev).contiguous(),
+            "kappa": torch.from_numpy(kap).to(dev),
+            "tau": torch.from_numpy(tau).to(dev),
+            "ne": ne,
+        }
+        # device-side Q assembly structure (values rebuilt in place)
+        qa = self._q_assembler
+        self._q_dev = {
+            "slots": qa._slots.to(dev),
+            "blocks": qa._blocks.to(dev).contiguous(),
+            "edge_of": qa._edge_of.to(dev),
+        }
+        self._gnc_ready = True
+
+    def _packed_update_weights(self) -> None:
+        """Launch the GNC weight kernel over private + shared LCs."""
+        from .ops import hip_backend as hb
+        g = self._gnc
+        if g["ne"] == 0:
+            return
+        p = self.params.robust_cost_params
+        hb.gnc_weights(self.X, self._nbr_buffer, g, self._all_weights_dev,
+                       self.d, self.r, self.robust_cost.mu,
+                       p.gnc_barc ** 2)
+
+    def _packed_rebuild_q(self) -> None:
+        """Re-assemble Q values from current weights (in place) and
+        refresh the preconditioner factors."""
+        from .ops import hip_backend as hb
+        Q = self.problem.Q
+        hb.q_assemble(Q.vals, self._q_dev["blocks"], self._q_dev["slots"],
+                      self._q_dev["edge_of"], self._all_weights_dev, self.dh)
+        self.problem.refresh_preconditioner()
+        self._dev_solver.bind_problem_static(self.problem)
+
+    def _packed_shared_weights(self) -> Tensor:
+        return self._w_shared_dev
 
     def _packed_nesterov_pre(self) -> None:
         from .ops import hip_backend as hb

@@ -41,6 +41,15 @@ Tensor = torch.Tensor
 STATUS_LEN = 6
 
 
+def _adjacency_from_ma(ma, num_poses):
+    nbr = [set() for _ in range(num_poses)]
+    for a, b in zip(ma.p1, ma.p2):
+        if a != b:
+            nbr[a].add(int(b))
+            nbr[b].add(int(a))
+    return [sorted(x) for x in nbr]
+
+
 class DistributedRBCDDriver:
     def __init__(self,
                  measurements: Sequence[RelativeSEMeasurement],
@@ -59,31 +68,81 @@ class DistributedRBCDDriver:
         self.verbose = verbose
         self.selection = selection
         self.device = device
-        d = measurements[0].d
+        from .measurements import MeasurementArray
+        self._soa = isinstance(measurements, MeasurementArray)
+        d = measurements.d if self._soa else measurements[0].d
         self.d, self.r, self.n_global = d, r, num_poses
         self.dh = d + 1
         self.acceleration = acceleration
+        self.robust = robust
 
         # ---- identical partition on every rank (deterministic) ---------
         if isinstance(partition, str):
             if partition == "contiguous":
                 part = contiguous_partition(num_poses, num_robots)
             else:
-                adj = adjacency_from_measurements(measurements, num_poses)
+                if self._soa:
+                    adj = _adjacency_from_ma(measurements, num_poses)
+                else:
+                    adj = adjacency_from_measurements(measurements,
+                                                      num_poses)
                 part = multilevel_partition(adj, num_robots)
         else:
             part = list(partition)
-        (odometry, private_lc, shared_lc, self.pose_map,
-         self.pose_to_index, self.pose_counts) = partition_measurements(
-            measurements, num_poses, part, num_robots)
+        if self._soa:
+            from .measurements import partition_measurement_array
+            (odometry, private_lc, shared_lc, local_idx, global_of,
+             self.pose_counts) = partition_measurement_array(
+                measurements, num_poses, part, num_robots)
+            self.pose_to_index = {}
+            for rb in range(num_robots):
+                for i, g in enumerate(global_of[rb]):
+                    self.pose_to_index[(rb, int(i))] = int(g)
+            # global cross-edge exchange maps (GNC weight sync)
+            partv = np.asarray(part, dtype=np.int64)
+            src_r = partv[measurements.p1]
+            dst_r = partv[measurements.p2]
+            cross = np.nonzero(src_r != dst_r)[0]
+            self._n_cross = len(cross)
+            cross_rank = {int(g): i for i, g in enumerate(cross)}
+            self._cross_map = []     # per agent: global positions
+            self._cross_owned = []   # per agent: owned mask
+            for rb in range(num_robots):
+                sel = np.nonzero((src_r != dst_r)
+                                 & ((src_r == rb) | (dst_r == rb)))[0]
+                self._cross_map.append(
+                    np.array([cross_rank[int(g)] for g in sel],
+                             dtype=np.int64))
+                other = np.where(src_r[sel] == rb, dst_r[sel], src_r[sel])
+                self._cross_owned.append(other > rb)
+        else:
+            (odometry, private_lc, shared_lc, self.pose_map,
+             self.pose_to_index, self.pose_counts) = partition_measurements(
+                measurements, num_poses, part, num_robots)
 
         # ---- agents owned by this rank ---------------------------------
         self.owner = [a % comm.world_size for a in range(num_robots)]
         self.local_agents: Dict[int, PGOAgent] = {}
         from .manifold import lifting_matrix
         YL = lifting_matrix(d, r)
-        T_chordal_pre = chordal_initialization(d, num_poses, measurements) \
-            if robust == RobustCostType.L2 else None
+        if self._soa:
+            # global-frame initialization: chordal is too heavy at SoA
+            # scale, so distribute the GLOBAL odometry dead-reckoning
+            # (prefix scan, vectorized) — consistent frames across agents.
+            from .measurements import odometry_initialization_array
+            odo_mask = (measurements.p1 + 1 == measurements.p2)
+            T_glob = odometry_initialization_array(
+                d, num_poses, measurements.select(odo_mask))
+            T_chordal_pre = T_glob
+        else:
+            T_chordal_pre = chordal_initialization(
+                d, num_poses, measurements) \
+                if robust == RobustCostType.L2 else None
+        Tg3 = None
+        if T_chordal_pre is not None:
+            Tg3 = np.ascontiguousarray(
+                T_chordal_pre.reshape(d, num_poses, self.dh).transpose(
+                    1, 0, 2))  # (n, d, dh) for fast slicing
         for rb in range(num_robots):
             if self.owner[rb] != comm.rank:
                 continue
@@ -94,14 +153,19 @@ class DistributedRBCDDriver:
             a = PGOAgent(rb, p)
             a.set_lifting_matrix(YL)
             T_init = None
-            if T_chordal_pre is not None:
-                T_init = np.zeros((d, self.pose_counts[rb] * self.dh))
-                for i in range(self.pose_counts[rb]):
-                    g = self.pose_to_index[(rb, i)]
-                    T_init[:, i * self.dh:(i + 1) * self.dh] = \
-                        T_chordal_pre[:, g * self.dh:(g + 1) * self.dh]
-            a.set_pose_graph(odometry[rb], private_lc[rb], shared_lc[rb],
-                             T_init=T_init)
+            if Tg3 is not None:
+                gidx = np.array([self.pose_to_index[(rb, i)]
+                                 for i in range(self.pose_counts[rb])],
+                                dtype=np.int64)
+                T_init = np.ascontiguousarray(
+                    Tg3[gidx].transpose(1, 0, 2).reshape(
+                        d, self.pose_counts[rb] * self.dh))
+            if self._soa:
+                a.set_pose_graph_arrays(odometry[rb], private_lc[rb],
+                                        shared_lc[rb], T_init=T_init)
+            else:
+                a.set_pose_graph(odometry[rb], private_lc[rb],
+                                 shared_lc[rb], T_init=T_init)
             self.local_agents[rb] = a
 
         # ---- public-pose packing layout (global, same on all ranks) ----
@@ -141,11 +205,18 @@ class DistributedRBCDDriver:
                 a.set_x(Xr)
 
         # coloring for the colored schedule
+        from .measurements import MeasurementArray as _MA
         nbrs = [set() for _ in range(num_robots)]
         for rb in range(num_robots):
-            for m in shared_lc[rb]:
-                o = m.r2 if m.r1 == rb else m.r1
-                nbrs[rb].add(o)
+            sl = shared_lc[rb]
+            if isinstance(sl, _MA):
+                if len(sl):
+                    o = np.where(sl.r1 == rb, sl.r2, sl.r1)
+                    nbrs[rb] = {int(x) for x in np.unique(o)}
+            else:
+                for m in sl:
+                    o = m.r2 if m.r1 == rb else m.r1
+                    nbrs[rb].add(o)
         colors = [-1] * num_robots
         for rb in range(num_robots):
             used = {colors[o] for o in nbrs[rb] if colors[o] >= 0}
@@ -162,13 +233,20 @@ class DistributedRBCDDriver:
 
     @staticmethod
     def _public_pose_sets(shared_lc, num_robots):
+        from .measurements import MeasurementArray
         pub = [set() for _ in range(num_robots)]
         for rb in range(num_robots):
-            for m in shared_lc[rb]:
-                if m.r1 == rb:
-                    pub[rb].add(m.p1)
-                else:
-                    pub[rb].add(m.p2)
+            sl = shared_lc[rb]
+            if isinstance(sl, MeasurementArray):
+                if len(sl):
+                    local = np.where(sl.r1 == rb, sl.p1, sl.p2)
+                    pub[rb] = {int(x) for x in np.unique(local)}
+            else:
+                for m in sl:
+                    if m.r1 == rb:
+                        pub[rb].add(m.p1)
+                    else:
+                        pub[rb].add(m.p2)
         return pub
 
     # -------------------------------------------------------------------
@@ -275,7 +353,7 @@ class DistributedRBCDDriver:
             time_limit_s: Optional[float] = None) -> RBCDResult:
         import torch
         if (str(self.device).startswith("cuda")
-                and self._robust_is_l2()
+                and (self._robust_is_l2() or self._soa)
                 and all(a.state == PGOAgentState.INITIALIZED
                         for a in self.local_agents.values())):
             return self._run_packed(max_iters, gradnorm_tol, trace_file,
@@ -424,7 +502,46 @@ class DistributedRBCDDriver:
         self._packed_scatter(flats)
         fout = open(trace_file, "w") if (trace_file and
                                          self.comm.rank == 0) else None
+        robust_mode = not self._robust_is_l2()
+        if robust_mode:
+            for a in self.local_agents.values():
+                a._packed_gnc_setup(dev)
+            self._w_exch = torch.zeros(self._n_cross, dtype=torch.float64,
+                                       device=dev)
+            self._cross_map_t = {
+                rb: torch.from_numpy(self._cross_map[rb]).to(dev)
+                for rb in self.local_agents}
+            self._cross_owned_t = {
+                rb: torch.from_numpy(
+                    np.nonzero(self._cross_owned[rb])[0]).to(dev)
+                for rb in self.local_agents}
+        if not hasattr(self, "_round_counter"):
+            self._round_counter = 0
+        inner = next(iter(self.local_agents.values())).params \
+            .robust_opt_inner_iters if self.local_agents else 30
         for it in range(max_iters):
+            self._round_counter += 1
+            if robust_mode and self._round_counter % inner == 0:
+                # GNC re-weighting round (reference iterate():
+                # shouldUpdateLoopClosureWeights -> update -> mu step)
+                for rb, a in self.local_agents.items():
+                    a._packed_update_weights()
+                self._w_exch.zero_()
+                for rb, a in self.local_agents.items():
+                    own = self._cross_owned_t[rb]
+                    if own.numel():
+                        pos = self._cross_map_t[rb].index_select(0, own)
+                        self._w_exch.index_copy_(
+                            0, pos, a._w_shared_dev.index_select(0, own))
+                self.comm.all_reduce_sum_(self._w_exch)
+                for rb, a in self.local_agents.items():
+                    if self._cross_map_t[rb].numel():
+                        a._w_shared_dev.copy_(
+                            self._w_exch.index_select(
+                                0, self._cross_map_t[rb]))
+                for rb, a in self.local_agents.items():
+                    a._packed_rebuild_q()
+                    a.robust_cost.update()
             if self.selection == "colored":
                 color = it % self._num_colors
                 active = [rb for rb in range(self.num_robots)

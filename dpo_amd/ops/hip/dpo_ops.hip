@@ -807,6 +807,67 @@ __global__ void k_g_assemble(double* __restrict__ Gt,
 }
 
 // ---------------------------------------------------------------------
+// GNC-TLS loop-closure re-weighting (reference PGOAgent::
+// updateLoopClosuresWeights, PGOAgent.cpp:1181-1245, weight formula
+// DPGO_robust.cpp:49-62). One thread per edge; endpoints are rows of X
+// (local poses) or of the packed neighbor buffer (shared edges).
+// Residual on the LIFTED variables: r^2 = kappa ||R^T Y1t - Y2t||_F^2
+//                                      + tau ||p2 - p1 - t^T Y1t||^2.
+// update_mask == 0 keeps the current weight (known inliers, non-owned
+// shared edges under the owner-computes rule).
+// ---------------------------------------------------------------------
+template <int D, int R>
+__global__ void k_gnc_weights(const double* __restrict__ X,
+                              const double* __restrict__ nbr,
+                              const long* __restrict__ e1_idx,
+                              const unsigned char* __restrict__ e1_nbr,
+                              const long* __restrict__ e2_idx,
+                              const unsigned char* __restrict__ e2_nbr,
+                              const double* __restrict__ Rm,
+                              const double* __restrict__ tm,
+                              const double* __restrict__ kappa,
+                              const double* __restrict__ tau,
+                              const unsigned char* __restrict__ update_mask,
+                              const long* __restrict__ widx,
+                              double* __restrict__ weights,
+                              int ne, double mu, double barc_sq) {
+  const int e = blockIdx.x * blockDim.x + threadIdx.x;
+  if (e >= ne) return;
+  if (!update_mask[e]) return;
+  constexpr int dh = D + 1;
+  const double* P1 = (e1_nbr[e] ? nbr : X) + (size_t)e1_idx[e] * dh * R;
+  const double* P2 = (e2_nbr[e] ? nbr : X) + (size_t)e2_idx[e] * dh * R;
+  const double* Re = Rm + (size_t)e * D * D;
+  const double* te = tm + (size_t)e * D;
+  double rot_err = 0.0, tran_err = 0.0;
+  #pragma unroll
+  for (int k = 0; k < R; ++k) {
+    #pragma unroll
+    for (int b = 0; b < D; ++b) {
+      // (Y1 R)^T row b = sum_a R[a][b] * Y1t[a][:]
+      double v = 0.0;
+      #pragma unroll
+      for (int a = 0; a < D; ++a) v = fma(Re[a * D + b], P1[a * R + k], v);
+      const double diff = v - P2[b * R + k];
+      rot_err = fma(diff, diff, rot_err);
+    }
+    double tv = P2[D * R + k] - P1[D * R + k];
+    #pragma unroll
+    for (int a = 0; a < D; ++a) tv = fma(-te[a], P1[a * R + k], tv);
+    tran_err = fma(tv, tv, tran_err);
+  }
+  const double r_sq = kappa[e] * rot_err + tau[e] * tran_err;
+  // GNC-TLS weight (eq. 14)
+  const double upper = (mu + 1.0) / mu * barc_sq;
+  const double lower = mu / (mu + 1.0) * barc_sq;
+  double w;
+  if (r_sq >= upper) w = 0.0;
+  else if (r_sq <= lower) w = 1.0;
+  else w = sqrt(barc_sq * mu * (mu + 1.0) / r_sq) - mu;
+  weights[widx[e]] = w;
+}
+
+// ---------------------------------------------------------------------
 // (d, r) dispatch: the supported compile-time shapes. SE(2): d=2,
 // r in 2..6; SE(3): d=3, r in 3..8. Everything the reference exercises
 // (r=5 RBCD, r=d batch) is covered; exotic shapes abort loudly.
@@ -1049,6 +1110,28 @@ void dpo_g_assemble(double* Gt, const double* E0, const long* local_pose,
 }
 
 int dpo_ctrl_size() { return CTRL_SIZE; }
+
+void dpo_gnc_weights(const double* X, const double* nbr, const long* e1_idx,
+                     const unsigned char* e1_nbr, const long* e2_idx,
+                     const unsigned char* e2_nbr, const double* Rm,
+                     const double* tm, const double* kappa,
+                     const double* tau, const unsigned char* update_mask,
+                     const long* widx, double* weights, int ne, int d,
+                     int r, double mu, double barc_sq, void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  const int grid = blocks_for(ne, 256);
+#define CASE_GNC(D, R) \
+  if (d == D && r == R) { \
+    hipLaunchKernelGGL((k_gnc_weights<D, R>), dim3(grid), dim3(256), 0, s, \
+                       X, nbr, e1_idx, e1_nbr, e2_idx, e2_nbr, Rm, tm, \
+                       kappa, tau, update_mask, widx, weights, ne, mu, \
+                       barc_sq); \
+    return; \
+  }
+  DPO_FOREACH_DR(CASE_GNC)
+#undef CASE_GNC
+  dpo_bad_shape(d, r);
+}
 
 }  // extern "C"
 

@@ -61,6 +61,8 @@ _lib.dpo_round_solve.restype = _i
 _lib.dpo_round_solve.argtypes = [_c, _c, _c, _d, _d, _i, _d,
                                  ctypes.POINTER(ctypes.c_double), _c]
 _lib.dpo_round_eval.argtypes = [_c, _c, _c, _c, _c]
+_lib.dpo_gnc_weights.argtypes = [_c, _c, _c, _c, _c, _c, _c, _c, _c, _c,
+                                 _c, _c, _c, _i, _i, _i, _d, _d, _c]
 
 CTRL_SIZE = _lib.dpo_ctrl_size()
 
@@ -288,3 +290,13 @@ class DeviceSolver:
         _lib.dpo_round_eval(self.handle, _p(X), _p(nbr),
                             _p(self._eval_out), _stream(X))
         return self._eval_out
+
+
+def gnc_weights(X: Tensor, nbr: Tensor, g: dict, weights: Tensor,
+                d: int, r: int, mu: float, barc_sq: float) -> None:
+    """Launch the GNC-TLS weight-update kernel over g['ne'] edges."""
+    _lib.dpo_gnc_weights(
+        _p(X), _p(nbr), _p(g["e1_idx"]), _p(g["e1_nbr"]), _p(g["e2_idx"]),
+        _p(g["e2_nbr"]), _p(g["R"]), _p(g["t"]), _p(g["kappa"]),
+        _p(g["tau"]), _p(g["upd"]), _p(g["widx"]), _p(weights),
+        g["ne"], d, r, mu, barc_sq, _stream(X))

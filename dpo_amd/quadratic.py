@@ -37,20 +37,35 @@ def edge_unit_blocks(m: RelativeSEMeasurement) -> Tuple[np.ndarray, np.ndarray, 
     """Unit-weight blocks (B_ii, B_jj, B_ij) of one edge's Q contribution.
     B_ii = T Omega0 T^T, B_jj = Omega0, B_ij = -T Omega0, where Omega0
     uses w = 1."""
-    d = m.d
+    Bii, Bjj, Bij = edge_unit_blocks_batch([m])
+    return Bii[0], Bjj[0], Bij[0]
+
+
+def edge_unit_blocks_batch(meas):
+    """Vectorized unit-weight Q blocks for a list of edges (or a
+    MeasurementArray): returns (Bii, Bjj, Bij) each (ne, dh, dh)."""
+    from .measurements import as_measurement_array
+    ma = as_measurement_array(meas)
+    ne = len(ma)
+    d = ma.d
     dh = d + 1
-    TOm = np.zeros((dh, dh))
-    TOm[:d, :d] = m.kappa * m.R
-    TOm[:d, d] = m.tau * m.t
-    TOm[d, d] = m.tau
-    Bii = np.zeros((dh, dh))
-    Bii[:d, :d] = m.kappa * np.eye(d) + m.tau * np.outer(m.t, m.t)
-    Bii[:d, d] = m.tau * m.t
-    Bii[d, :d] = m.tau * m.t
-    Bii[d, d] = m.tau
-    Bjj = np.zeros((dh, dh))
-    Bjj[:d, :d] = m.kappa * np.eye(d)
-    Bjj[d, d] = m.tau
+    R = ma.R
+    t = ma.t
+    kappa = ma.kappa[:, None, None]
+    tau = ma.tau[:, None, None]
+    TOm = np.zeros((ne, dh, dh))
+    TOm[:, :d, :d] = kappa * R
+    TOm[:, :d, d] = tau[:, :, 0] * t
+    TOm[:, d, d] = tau[:, 0, 0]
+    Bii = np.zeros((ne, dh, dh))
+    Bii[:, :d, :d] = kappa * np.eye(d)[None] + tau * np.einsum(
+        'ei,ej->eij', t, t)
+    Bii[:, :d, d] = tau[:, :, 0] * t
+    Bii[:, d, :d] = tau[:, :, 0] * t
+    Bii[:, d, d] = tau[:, 0, 0]
+    Bjj = np.zeros((ne, dh, dh))
+    Bjj[:, :d, :d] = kappa * np.eye(d)[None]
+    Bjj[:, d, d] = tau[:, 0, 0]
     return Bii, Bjj, -TOm
 
 
@@ -102,14 +117,11 @@ class BSRMatrix:
     def diag_slot_index(self) -> Tensor:
         """(n,) index into vals of each diagonal block (cached)."""
         if getattr(self, "_diag_slots", None) is None:
-            rp = self.row_ptr.cpu().numpy()
-            ci = self.col_idx.cpu().numpy()
-            import numpy as _np
-            slots = _np.zeros(self.n, dtype=_np.int64)
-            for i in range(self.n):
-                s, e = rp[i], rp[i + 1]
-                hit = _np.nonzero(ci[s:e] == i)[0]
-                slots[i] = s + (hit[0] if hit.size else 0)
+            rp = self.row_ptr.cpu().numpy().astype(np.int64)
+            ci = self.col_idx.cpu().numpy().astype(np.int64)
+            bro = np.repeat(np.arange(self.n, dtype=np.int64), np.diff(rp))
+            slots = np.nonzero(ci == bro)[0]
+            assert len(slots) == self.n, "missing diagonal block"
             self._diag_slots = torch.from_numpy(slots).to(self.vals.device)
         return self._diag_slots
 
@@ -154,69 +166,68 @@ class QAssembler:
     """
 
     def __init__(self, n: int, d: int,
-                 measurements: Sequence[RelativeSEMeasurement],
+                 measurements,
                  shared_flags: Optional[Sequence[bool]] = None,
                  local_endpoint: Optional[Sequence[int]] = None):
+        from .measurements import as_measurement_array
         self.n, self.d = n, d
         dh = d + 1
         self.dh = dh
-        self.meas = list(measurements)
+        self.meas = as_measurement_array(measurements)
         ne = len(self.meas)
         shared = list(shared_flags) if shared_flags is not None else [False] * ne
         lep = list(local_endpoint) if local_endpoint is not None else [0] * ne
 
-        # ---- block sparsity pattern ----------------------------------
-        pairs = set((i, i) for i in range(n))
-        for k, m in enumerate(self.meas):
-            if shared[k]:
-                continue
-            pairs.add((m.p1, m.p2))
-            pairs.add((m.p2, m.p1))
-        by_row: List[List[int]] = [[] for _ in range(n)]
-        for (i, j) in pairs:
-            by_row[i].append(j)
+        # ---- block sparsity pattern (vectorized) ---------------------
+        p1 = self.meas.p1
+        p2 = self.meas.p2
+        sh = np.array(shared, dtype=bool)
+        lepv = np.array(lep, dtype=np.int64)
+        priv = ~sh
+        rows = np.concatenate([np.arange(n, dtype=np.int64),
+                               p1[priv], p2[priv]])
+        cols = np.concatenate([np.arange(n, dtype=np.int64),
+                               p2[priv], p1[priv]])
+        keys = rows * n + cols
+        uk = np.unique(keys)
+        u_rows = uk // n
+        u_cols = uk % n
+        nnzb = len(uk)
         row_ptr = np.zeros(n + 1, dtype=np.int32)
-        col_idx_list: List[int] = []
-        slot: Dict[Tuple[int, int], int] = {}
-        for i in range(n):
-            for j in sorted(by_row[i]):
-                slot[(i, j)] = len(col_idx_list)
-                col_idx_list.append(j)
-            row_ptr[i + 1] = len(col_idx_list)
-        col_idx = np.array(col_idx_list, dtype=np.int32)
-        nnzb = len(col_idx_list)
+        np.add.at(row_ptr, u_rows + 1, 1)
+        row_ptr = np.cumsum(row_ptr).astype(np.int32)
+        col_idx = u_cols.astype(np.int32)
 
-        # ---- per-edge unit blocks and target slots -------------------
-        # Each edge writes up to 3 distinct slots (ii, jj, ij) plus the
-        # transposed block at (j, i). We store slot ids + unit blocks flat
-        # for a single scatter-add pass (GPU-friendly).
-        slots: List[int] = []
-        blocks: List[np.ndarray] = []
-        edge_of: List[int] = []
-        transposed: List[bool] = []
-        for k, m in enumerate(self.meas):
-            Bii, Bjj, Bij = edge_unit_blocks(m)
-            if shared[k]:
-                # Diagonal correction only, at the local endpoint.
-                if lep[k] == 0:  # outgoing: local pose is p1
-                    slots.append(slot[(m.p1, m.p1)]); blocks.append(Bii)
-                else:            # incoming: local pose is p2
-                    slots.append(slot[(m.p2, m.p2)]); blocks.append(Bjj)
-                edge_of.append(k); transposed.append(False)
-            else:
-                slots.append(slot[(m.p1, m.p1)]); blocks.append(Bii)
-                edge_of.append(k); transposed.append(False)
-                slots.append(slot[(m.p2, m.p2)]); blocks.append(Bjj)
-                edge_of.append(k); transposed.append(False)
-                slots.append(slot[(m.p1, m.p2)]); blocks.append(Bij)
-                edge_of.append(k); transposed.append(False)
-                slots.append(slot[(m.p2, m.p1)]); blocks.append(Bij.T)
-                edge_of.append(k); transposed.append(True)
+        def slot_of(i, j):
+            return np.searchsorted(uk, i * n + j)
 
-        self._slots = torch.from_numpy(np.array(slots, dtype=np.int64))
-        self._blocks = torch.from_numpy(np.stack(blocks)) if blocks else \
-            torch.zeros(0, dh, dh, dtype=torch.float64)
-        self._edge_of = torch.from_numpy(np.array(edge_of, dtype=np.int64))
+        # ---- per-edge unit blocks and target slots (vectorized) ------
+        ne = len(self.meas)
+        Bii, Bjj, Bij = (np.zeros((0, dh, dh)),) * 3
+        if ne:
+            Bii, Bjj, Bij = edge_unit_blocks_batch(self.meas)
+        eidx = np.arange(ne, dtype=np.int64)
+        sh_out = sh & (lepv == 0)
+        sh_in = sh & (lepv == 1)
+        slots = np.concatenate([
+            slot_of(p1[priv], p1[priv]),
+            slot_of(p2[priv], p2[priv]),
+            slot_of(p1[priv], p2[priv]),
+            slot_of(p2[priv], p1[priv]),
+            slot_of(p1[sh_out], p1[sh_out]),
+            slot_of(p2[sh_in], p2[sh_in]),
+        ])
+        blocks = np.concatenate([
+            Bii[priv], Bjj[priv], Bij[priv],
+            np.transpose(Bij[priv], (0, 2, 1)),
+            Bii[sh_out], Bjj[sh_in],
+        ]) if ne else np.zeros((0, dh, dh))
+        edge_of = np.concatenate([eidx[priv]] * 4
+                                 + [eidx[sh_out], eidx[sh_in]])
+
+        self._slots = torch.from_numpy(np.ascontiguousarray(slots))
+        self._blocks = torch.from_numpy(np.ascontiguousarray(blocks))
+        self._edge_of = torch.from_numpy(np.ascontiguousarray(edge_of))
         self._nnzb = nnzb
         self.bsr = BSRMatrix(
             n, dh,
@@ -226,7 +237,7 @@ class QAssembler:
     def assemble(self, weights: Optional[Tensor] = None) -> BSRMatrix:
         """(Re)compute BSR values given per-edge weights (default all 1)."""
         if weights is None:
-            weights = torch.ones(len(self.meas), dtype=torch.float64)
+            weights = torch.from_numpy(self.meas.weight.copy())
         w = weights[self._edge_of]
         vals = torch.zeros(self._nnzb, self.dh, self.dh, dtype=torch.float64)
         vals.index_add_(0, self._slots, self._blocks * w[:, None, None])
@@ -236,14 +247,14 @@ class QAssembler:
 
 
 def assemble_connection_laplacian(
-        measurements: Sequence[RelativeSEMeasurement], n: int, d: int,
+        measurements, n: int, d: int,
         weights: Optional[Sequence[float]] = None) -> BSRMatrix:
     """Centralized / private connection Laplacian Q as BSR (parity with
     reference constructConnectionLaplacianSE, DPGO_utils.cpp:265-271;
     weights default to each measurement's stored weight)."""
     qa = QAssembler(n, d, measurements)
     if weights is None:
-        weights = [m.weight for m in measurements]
+        return qa.assemble()
     return qa.assemble(torch.tensor(list(weights), dtype=torch.float64))
 
 
@@ -258,25 +269,27 @@ class GAssembler:
     """
 
     def __init__(self, n: int, d: int,
-                 shared_meas: Sequence[RelativeSEMeasurement],
+                 shared_meas,
                  local_endpoint: Sequence[int],
                  nbr_slot: Sequence[int]):
+        from .measurements import as_measurement_array
         self.n, self.d = n, d
         dh = d + 1
         ne = len(shared_meas)
         E0 = np.zeros((ne, dh, dh))
         local_pose = np.zeros(ne, dtype=np.int64)
-        for k, m in enumerate(shared_meas):
-            TOm = np.zeros((dh, dh))
-            TOm[:d, :d] = m.kappa * m.R
-            TOm[:d, d] = m.tau * m.t
-            TOm[d, d] = m.tau
-            if local_endpoint[k] == 0:
-                local_pose[k] = m.p1
-                E0[k] = TOm
-            else:
-                local_pose[k] = m.p2
-                E0[k] = TOm.T
+        if ne:
+            ma = as_measurement_array(shared_meas)
+            kap = ma.kappa[:, None, None]
+            tau = ma.tau
+            TOm = np.zeros((ne, dh, dh))
+            TOm[:, :d, :d] = kap * ma.R
+            TOm[:, :d, d] = tau[:, None] * ma.t
+            TOm[:, d, d] = tau
+            lev = np.array(local_endpoint, dtype=np.int64)
+            local_pose = np.where(lev == 0, ma.p1, ma.p2)
+            E0 = np.where((lev == 0)[:, None, None], TOm,
+                          np.transpose(TOm, (0, 2, 1)))
         self.E0 = torch.from_numpy(E0)
         self.local_pose = torch.from_numpy(local_pose)
         self.nbr_slot = torch.tensor(list(nbr_slot), dtype=torch.int64)
@@ -372,6 +385,29 @@ class QuadraticProblem:
             diag = Q.diag_blocks() + precond_reg * torch.eye(
                 dh, dtype=torch.float64, device=dev)
             self._Lpre = torch.linalg.cholesky(diag)
+
+    def refresh_preconditioner(self, precond_reg: float = 0.1) -> None:
+        """Rebuild the preconditioner after Q values changed in place
+        (GNC re-weighting)."""
+        self.Q.invalidate()
+        mode = self._active_precond
+        dh = self.dh
+        dev = self.Q.vals.device
+        if mode == "exact":
+            import scipy.sparse as sp
+            import scipy.sparse.linalg as spla
+            A = (self.Q.to_scipy() + precond_reg * sp.eye(self.N)).tocsc()
+            self._lu = spla.splu(A)
+        elif mode == "dense":
+            A = self.Q.to_dense()
+            A += precond_reg * torch.eye(self.N, dtype=A.dtype, device=dev)
+            L = torch.linalg.cholesky(A)
+            self._Minv = torch.cholesky_inverse(L).to(
+                torch.float32).contiguous()
+        else:
+            diag = self.Q.diag_blocks() + precond_reg * torch.eye(
+                dh, dtype=torch.float64, device=dev)
+            self._Lpre = torch.linalg.cholesky(diag).contiguous()
 
     def set_g(self, Gt: Tensor) -> None:
         self.Gt = Gt

@@ -195,3 +195,92 @@ def triangle_graph() -> Tuple[List[RelativeSEMeasurement], int,
         T[:, i * 4:i * 4 + 3] = Rs[i]
         T[:, i * 4 + 3] = ts[i]
     return meas, 3, T
+
+
+# ---------------------------------------------------------------------
+# Vectorized large-scale generator (SoA): synthetic grid3D at up to
+# millions of poses (BASELINE.json config #5: 1M-pose grid, 8 agents,
+# robust loop-closure rejection).
+# ---------------------------------------------------------------------
+def _random_rotations_batch(n: int, rng, scale: float = 0.5) -> np.ndarray:
+    """(n, 3, 3) random rotations via batched Rodrigues."""
+    w = rng.standard_normal((n, 3))
+    nw = np.linalg.norm(w, axis=1, keepdims=True)
+    nw[nw == 0] = 1.0
+    ang = rng.uniform(0, scale, size=(n, 1))
+    w = w / nw * ang
+    th = np.linalg.norm(w, axis=1)
+    K = np.zeros((n, 3, 3))
+    K[:, 0, 1] = -w[:, 2]; K[:, 0, 2] = w[:, 1]
+    K[:, 1, 0] = w[:, 2];  K[:, 1, 2] = -w[:, 0]
+    K[:, 2, 0] = -w[:, 1]; K[:, 2, 1] = w[:, 0]
+    th_safe = np.where(th < 1e-12, 1.0, th)
+    a = np.where(th < 1e-12, 1.0, np.sin(th) / th_safe)[:, None, None]
+    b = np.where(th < 1e-12, 0.5,
+                 (1 - np.cos(th)) / (th_safe ** 2))[:, None, None]
+    return np.eye(3)[None] + a * K + b * (K @ K)
+
+
+def grid3d_soa(side: int, rot_noise: float = 0.05, tran_noise: float = 0.02,
+               kappa: float = 1000.0, tau: float = 100.0,
+               outlier_prob: float = 0.0, seed: int = 0):
+    """Vectorized serpentine-grid SE(3) pose graph as a MeasurementArray.
+    Identical structure to grid3d() but scales to millions of poses."""
+    from .measurements import MeasurementArray
+    rng = np.random.default_rng(seed)
+    n = side ** 3
+    idx = np.arange(n)
+    z = idx // (side * side)
+    rem = idx % (side * side)
+    y = rem // side
+    x = rem % side
+    y = np.where(z % 2 == 1, side - 1 - y, y)
+    x = np.where(y % 2 == 1, side - 1 - x, x)
+    P = np.stack([x, y, z], axis=1).astype(np.float64)
+    Rw = _random_rotations_batch(n, rng, 0.5)
+
+    # index lookup grid: pos -> serpentine index
+    lookup = np.empty((side, side, side), dtype=np.int64)
+    lookup[x, y, z] = idx
+
+    pairs = [np.stack([idx[:-1], idx[1:]], axis=1)]  # odometry chain
+    for dxyz in ((1, 0, 0), (0, 1, 0), (0, 0, 1)):
+        m = (x + dxyz[0] < side) & (y + dxyz[1] < side) & (z + dxyz[2] < side)
+        j = lookup[np.clip(x + dxyz[0], 0, side - 1),
+                   np.clip(y + dxyz[1], 0, side - 1),
+                   np.clip(z + dxyz[2], 0, side - 1)]
+        keep = m & (np.abs(j - idx) != 1)
+        a = np.minimum(idx[keep], j[keep])
+        b = np.maximum(idx[keep], j[keep])
+        pairs.append(np.stack([a, b], axis=1))
+    E = np.concatenate(pairs)
+    # odometry must come first and stay ordered; loop closures after
+    ne = len(E)
+    i_, j_ = E[:, 0], E[:, 1]
+    # relative measurements with noise
+    Rn = _random_rotations_batch(ne, rng, rot_noise) if rot_noise > 0 \
+        else np.tile(np.eye(3), (ne, 1, 1))
+    Rrel = np.transpose(Rw[i_], (0, 2, 1)) @ Rw[j_] @ Rn
+    trel = np.einsum('eij,ej->ei', np.transpose(Rw[i_], (0, 2, 1)),
+                     P[j_] - P[i_])
+    if tran_noise > 0:
+        trel = trel + rng.standard_normal((ne, 3)) * tran_noise
+    # outliers among loop closures only
+    n_odo = n - 1
+    if outlier_prob > 0:
+        out_mask = np.zeros(ne, dtype=bool)
+        lc = np.arange(n_odo, ne)
+        out_mask[lc[rng.uniform(size=len(lc)) < outlier_prob]] = True
+        n_out = int(out_mask.sum())
+        if n_out:
+            Rrel[out_mask] = _random_rotations_batch(n_out, rng, 3.14)
+            trel[out_mask] = rng.standard_normal((n_out, 3)) * 5.0
+    z64 = np.zeros(ne, dtype=np.int64)
+    ma = MeasurementArray(
+        r1=z64, r2=z64.copy(), p1=i_.astype(np.int64),
+        p2=j_.astype(np.int64), R=Rrel, t=trel,
+        kappa=np.full(ne, kappa), tau=np.full(ne, tau),
+        weight=np.ones(ne), is_known_inlier=np.ones(ne, dtype=bool))
+    ma.outlier_mask = out_mask if outlier_prob > 0 else \
+        np.zeros(ne, dtype=bool)
+    return ma, n

@@ -210,3 +210,71 @@ def test_dist_driver_packed_gpu_accelerated():
                                 acceleration=True)
     res = gpu.run(max_iters=250)
     assert res.converged
+
+
+def test_soa_robust_gnc_gpu():
+    """SoA pipeline + packed robust GNC on GPU: planted outlier loop
+    closures must be rejected (weight -> 0) and inliers kept."""
+    import numpy as np
+    from dpo_amd.comm import Comm
+    from dpo_amd.dist_driver import DistributedRBCDDriver
+    from dpo_amd.synthetic import grid3d_soa
+    from dpo_amd.types import RobustCostType
+    ma, n = grid3d_soa(side=8, outlier_prob=0.15, seed=3,
+                       rot_noise=0.02, tran_noise=0.01)
+    outliers = ma.outlier_mask.copy()
+    drv = DistributedRBCDDriver(
+        ma, n, 2, Comm(), r=5, partition="contiguous",
+        robust=RobustCostType.GNC_TLS, device=DEV)
+    # accelerated GNC schedule for the small fixture
+    for a in drv.local_agents.values():
+        a.params.robust_opt_inner_iters = 10
+        a.robust_cost.params.gnc_mu_step = 2.5
+        a.robust_cost.params.gnc_init_mu = 1e-3
+        a.robust_cost.reset()
+    res = drv.run(max_iters=400, gradnorm_tol=0.05)
+    # collect final weights mapped back to global edges
+    # (2 agents, contiguous: use each agent's arrays)
+    n_rej = n_rej_true = n_kept_inlier = n_inlier = 0
+    for rb, a in drv.local_agents.items():
+        w = a._all_weights_dev.cpu().numpy()
+        n_odo = len(a._odo_ma)
+        for k in range(len(a._priv_ma)):
+            wk = w[n_odo + k]
+            if wk < 0.5:
+                n_rej += 1
+        # crude: count kept weights
+        lcw = w[n_odo:]
+        n_inlier += int((lcw > 0.9).sum())
+        n_rej_true += int((lcw < 0.1).sum())
+    frac_out = outliers.mean()
+    total_lc = sum(len(a._priv_ma) + len(a._shared_ma)
+                   for a in drv.local_agents.values())
+    # rejected fraction should be in the ballpark of the planted fraction
+    assert n_rej_true > 0.5 * frac_out * total_lc, \
+        f"rejected {n_rej_true} of ~{frac_out * total_lc:.0f} planted"
+    assert n_inlier > 0.6 * total_lc
+    # cost should be far below the unweighted initial cost
+    assert res.trace[-1][0] < res.trace[0][0]
+
+
+def test_soa_l2_gpu_matches_object_path():
+    """SoA construction must produce the same optimization as the object
+    path on an L2 run."""
+    from dpo_amd.comm import Comm
+    from dpo_amd.dist_driver import DistributedRBCDDriver
+    from dpo_amd.synthetic import grid3d, grid3d_soa
+    from dpo_amd.measurements import MeasurementArray
+    meas, n = grid3d(side=4, seed=0)
+    ma = MeasurementArray.from_list(meas)
+    obj = DistributedRBCDDriver(meas, n, 2, Comm(), r=5,
+                                partition="contiguous", device=DEV)
+    res_obj = obj.run(max_iters=100)
+    soa = DistributedRBCDDriver(ma, n, 2, Comm(), r=5,
+                                partition="contiguous", device=DEV)
+    res_soa = soa.run(max_iters=100)
+    # different initializations (chordal vs odometry-prefix for L2? both
+    # use chordal path in object mode; SoA L2 also distributes odometry
+    # init) — compare converged cost instead of traces
+    assert abs(res_soa.final_cost - res_obj.final_cost) < 1e-2 * max(
+        1.0, abs(res_obj.final_cost))
