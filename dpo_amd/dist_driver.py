@@ -127,13 +127,16 @@ class DistributedRBCDDriver:
         YL = lifting_matrix(d, r)
         if self._soa:
             # global-frame initialization: chordal is too heavy at SoA
-            # scale, so distribute the GLOBAL odometry dead-reckoning
+            # scale, so distribute an externally provided warm start
+            # (ma.warm_start) or the GLOBAL odometry dead-reckoning
             # (prefix scan, vectorized) — consistent frames across agents.
-            from .measurements import odometry_initialization_array
-            odo_mask = (measurements.p1 + 1 == measurements.p2)
-            T_glob = odometry_initialization_array(
-                d, num_poses, measurements.select(odo_mask))
-            T_chordal_pre = T_glob
+            if getattr(measurements, "warm_start", None) is not None:
+                T_chordal_pre = measurements.warm_start
+            else:
+                from .measurements import odometry_initialization_array
+                odo_mask = (measurements.p1 + 1 == measurements.p2)
+                T_chordal_pre = odometry_initialization_array(
+                    d, num_poses, measurements.select(odo_mask))
         else:
             T_chordal_pre = chordal_initialization(
                 d, num_poses, measurements) \

@@ -283,4 +283,10 @@ def grid3d_soa(side: int, rot_noise: float = 0.05, tran_noise: float = 0.02,
         weight=np.ones(ne), is_known_inlier=np.ones(ne, dtype=bool))
     ma.outlier_mask = out_mask if outlier_prob > 0 else \
         np.zeros(ne, dtype=bool)
+    # stash ground truth (d, n*(d+1)) for warm-start experiments
+    T_gt = np.zeros((3, n * 4))
+    Tv = T_gt.reshape(3, n, 4).transpose(1, 0, 2)
+    Tv[:, :, :3] = Rw
+    Tv[:, :, 3] = P
+    ma.ground_truth = T_gt
     return ma, n
