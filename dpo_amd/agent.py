@@ -706,6 +706,23 @@ class PGOAgent:
             Ti[:, self.d] -= Ya.T @ pa
             return Ti
 
+    def get_neighbor_pose_in_global_frame(self, neighbor_id: int,
+                                          pose_id: int
+                                          ) -> Optional[np.ndarray]:
+        """Round a cached neighbor pose to SE(d) in the global frame
+        (reference PGOAgent.cpp:540-562)."""
+        if self.global_anchor is None or self.state != PGOAgentState.INITIALIZED:
+            return None
+        with self._lock:
+            v = self.neighbor_pose_dict.get((neighbor_id, pose_id))
+            if v is None:
+                return None
+            Ya = self.global_anchor[:, :self.d]
+            pa = self.global_anchor[:, self.d]
+            Ti = Ya.T @ v
+            Ti[:, self.d] -= Ya.T @ pa
+            return Ti
+
     def local_pose_graph_optimization(self) -> np.ndarray:
         """Single-robot full-batch RTR at r = d (reference
         PGOAgent.cpp:964-990; batch knob set at 981-984)."""

@@ -28,6 +28,9 @@ def main():
     ap.add_argument("--max-iters", type=int, default=1000)
     ap.add_argument("--tol", type=float, default=0.1)
     ap.add_argument("--trace", default=None)
+    ap.add_argument("--dump-trajectory", default=None,
+                    help="write the final rounded global trajectory as "
+                         "CSV (reference PartitionInitial.cpp:329-335)")
     ap.add_argument("--driver", default="dist", choices=["dist", "local"])
     args = ap.parse_args()
 
@@ -63,6 +66,13 @@ def main():
         "ms_per_iter": res.elapsed_s / max(res.iterations, 1) * 1e3,
     }
     rank = int(os.environ.get("RANK", "0"))
+    if args.dump_trajectory and args.driver == "local" and rank == 0:
+        import numpy as np
+        from dpo_amd.logger import PGOLogger
+        T = drv.final_trajectory()
+        lg = PGOLogger(os.path.dirname(args.dump_trajectory) or ".")
+        lg.log_trajectory(meas[0].d, n, T,
+                          os.path.basename(args.dump_trajectory))
     if rank == 0:
         print(json.dumps(out))
 
