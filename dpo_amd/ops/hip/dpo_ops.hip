@@ -26,6 +26,7 @@
 #include <math.h>
 #include <stdint.h>
 #include <stdio.h>
+#include <string.h>
 
 #define DPO_CHECK(x)                                                     \
   do {                                                                   \
@@ -1171,6 +1172,17 @@ struct DpoCtx {
   const double *g_w = nullptr;
   int g_ne = 0;
   double *G_buf = nullptr;  // ctx-owned (N, r) linear-term buffer
+  // hipGraph caches for the fixed-pointer round sequences. Keyed by the
+  // operand pointers; invalidated whenever set_problem/set_gdata change
+  // them (preconditioner refresh, Q rebuild keep the same buffers).
+  hipGraphExec_t solve_graph = nullptr;
+  const void* solve_key[4] = {};
+  hipGraphExec_t eval_graph = nullptr;
+  const void* eval_key[4] = {};
+  void invalidate_graphs() {
+    if (solve_graph) { hipGraphExecDestroy(solve_graph); solve_graph = nullptr; }
+    if (eval_graph) { hipGraphExecDestroy(eval_graph); eval_graph = nullptr; }
+  }
 };
 
 static void ctx_assemble_g(DpoCtx* c, const double* nbr, hipStream_t s) {
@@ -1228,6 +1240,7 @@ void dpo_ctx_destroy(void* h) {
   hipFree(c->W); hipFree(c->grad); hipFree(c->eta); hipFree(c->delta);
   hipFree(c->rvec); hipFree(c->z); hipFree(c->Hd); hipFree(c->step);
   hipFree(c->Xprop); hipFree(c->eta_snap); hipFree(c->delta_snap);
+  c->invalidate_graphs();
   hipFree(c->ctrl); hipFree(c->G_buf); hipHostFree(c->ctrl_host);
   delete c;
 }
@@ -1236,21 +1249,23 @@ void dpo_ctx_set_problem(void* h, const int* rp, const int* ci,
                          const double* vals, const double* Gt,
                          const float* Minv, const double* Ljac) {
   DpoCtx* c = (DpoCtx*)h;
+  if (c->q_vals != vals || c->Minv != Minv || c->Ljac != Ljac
+      || c->Gt != Gt || c->q_rp != rp)
+    c->invalidate_graphs();
   c->q_rp = rp; c->q_ci = ci; c->q_vals = vals;
   c->Gt = Gt; c->Minv = Minv; c->Ljac = Ljac;
 }
 
-// Full RBCD local solve in place on X. stats_out (host, >= 8 doubles):
-// [status, f_init, gn_init, f_opt, gn_opt, rho, shrink_count, iters]
-int dpo_rbcd_solve(void* h, double* X, double tol, double Delta0,
-                   int max_shrink, double accept_rho,
-                   int compute_final_gn, double* stats_out, void* stream) {
-  DpoCtx* c = (DpoCtx*)h;
-  hipStream_t s = (hipStream_t)stream;
+// Enqueue the fixed pre-sync solve sequence (gradient, tCG, first
+// candidate + acceptance test). Capturable as one hipGraph: every
+// operand pointer is stable across rounds by construction.
+static void enqueue_solve_body(DpoCtx* c, double* X, const double* nbr,
+                               double tol, double Delta0,
+                               double accept_rho, hipStream_t s) {
   const int n = c->n, d = c->d, r = c->r;
   const long total = c->total;
   const int gvec = (int)((total + 255) / 256);
-
+  if (nbr) ctx_assemble_g(c, nbr, s);
   DPO_CHECK(hipMemsetAsync(c->ctrl, 0, CTRL_SIZE * sizeof(double), s));
   DPO_CHECK(hipMemsetAsync(c->eta, 0, total * sizeof(double), s));
   DPO_CHECK(hipMemsetAsync(c->delta, 0, total * sizeof(double), s));
@@ -1294,28 +1309,91 @@ int dpo_rbcd_solve(void* h, double* X, double tol, double Delta0,
                        c->delta, c->z, c->ctrl, total);
   }
   hipLaunchKernelGGL(k_ctrl_tcg_end, dim3(1), dim3(64), 0, s, c->ctrl);
+  // first candidate attempt is part of the fixed sequence
+  hipLaunchKernelGGL(k_ctrl_candidate, dim3(1), dim3(64), 0, s, c->ctrl);
+  hipLaunchKernelGGL(k_form_step, dim3(gvec), dim3(256), 0, s,
+                     c->step, c->eta, c->eta_snap, c->delta_snap,
+                     c->ctrl, total);
+  launch_polar(X, c->step, nullptr, 1.0, 1.0, 0.0, c->Xprop, n, d, r,
+               c->ctrl, ST_TCG_STOP, s);
+  ctx_spmm(c, c->Xprop, c->W, ST_TCG_STOP, s);
+  hipLaunchKernelGGL(k_dots, dim3(gvec), dim3(256), 0, s,
+                     c->Xprop, c->W, c->Gt, c->ctrl, C_DOT0, C_DOT2,
+                     total, ST_TCG_STOP);
+  hipLaunchKernelGGL(k_ctrl_accept, dim3(1), dim3(64), 0, s, c->ctrl,
+                     accept_rho);
+  DPO_CHECK(hipMemcpyAsync(c->ctrl_host, c->ctrl,
+                           CTRL_SIZE * sizeof(double),
+                           hipMemcpyDeviceToHost, s));
+}
 
-  // candidate / shrink loop (host loop, one sync per attempt; the
-  // accepted-first-try case costs exactly one sync)
+// Full RBCD local solve in place on X. stats_out (host, >= 8 doubles):
+// [status, f_init, gn_init, f_opt, gn_opt, rho, shrink_count, iters]
+// nbr != null: assemble G from the packed neighbor buffer first; the
+// whole pre-sync sequence is then served from a cached hipGraph.
+static int rbcd_solve_impl(DpoCtx* c, double* X, const double* nbr,
+                           double tol, double Delta0, int max_shrink,
+                           double accept_rho, int compute_final_gn,
+                           double* stats_out, hipStream_t s) {
+  const int n = c->n, d = c->d, r = c->r;
+  const long total = c->total;
+  const int gvec = (int)((total + 255) / 256);
+
+  const void* key[4] = {X, nbr, (const void*)(intptr_t)(tol * 1e9),
+                        (const void*)(intptr_t)Delta0};
+  bool key_match = c->solve_graph && memcmp(key, c->solve_key,
+                                            sizeof(key)) == 0;
+  if (!key_match) {
+    c->invalidate_graphs();
+    hipGraph_t graph = nullptr;
+    hipError_t rc = hipStreamBeginCapture(
+        s, hipStreamCaptureModeThreadLocal);
+    if (rc == hipSuccess) {
+      enqueue_solve_body(c, X, nbr, tol, Delta0, accept_rho, s);
+      rc = hipStreamEndCapture(s, &graph);
+    }
+    if (rc == hipSuccess && graph) {
+      rc = hipGraphInstantiate(&c->solve_graph, graph, nullptr, nullptr, 0);
+      hipGraphDestroy(graph);
+    }
+    if (rc != hipSuccess || !c->solve_graph) {
+      // capture unavailable: run eagerly
+      c->solve_graph = nullptr;
+      enqueue_solve_body(c, X, nbr, tol, Delta0, accept_rho, s);
+      DPO_CHECK(hipStreamSynchronize(s));
+    } else {
+      memcpy(c->solve_key, key, sizeof(key));
+    }
+  }
+  if (c->solve_graph) {
+    DPO_CHECK(hipGraphLaunch(c->solve_graph, s));
+    DPO_CHECK(hipStreamSynchronize(s));
+  }
+
+  // shrink loop (rare: only after a rejected first attempt)
   int status = ST_GIVE_UP;
   int shrinks = 0;
   for (int attempt = 0; attempt <= max_shrink; ++attempt) {
-    hipLaunchKernelGGL(k_ctrl_candidate, dim3(1), dim3(64), 0, s, c->ctrl);
-    hipLaunchKernelGGL(k_form_step, dim3(gvec), dim3(256), 0, s,
-                       c->step, c->eta, c->eta_snap, c->delta_snap,
-                       c->ctrl, total);
-    launch_polar(X, c->step, nullptr, 1.0, 1.0, 0.0, c->Xprop, n, d, r,
-                 c->ctrl, ST_TCG_STOP, s);
-    ctx_spmm(c, c->Xprop, c->W, ST_TCG_STOP, s);
-    hipLaunchKernelGGL(k_dots, dim3(gvec), dim3(256), 0, s,
-                       c->Xprop, c->W, c->Gt, c->ctrl, C_DOT0, C_DOT2,
-                       total, ST_TCG_STOP);
-    hipLaunchKernelGGL(k_ctrl_accept, dim3(1), dim3(64), 0, s, c->ctrl,
-                       accept_rho);
-    DPO_CHECK(hipMemcpyAsync(c->ctrl_host, c->ctrl,
-                             CTRL_SIZE * sizeof(double),
-                             hipMemcpyDeviceToHost, s));
-    DPO_CHECK(hipStreamSynchronize(s));
+    if (attempt > 0) {
+      hipLaunchKernelGGL(k_ctrl_shrink, dim3(1), dim3(64), 0, s, c->ctrl);
+      hipLaunchKernelGGL(k_ctrl_candidate, dim3(1), dim3(64), 0, s,
+                         c->ctrl);
+      hipLaunchKernelGGL(k_form_step, dim3(gvec), dim3(256), 0, s,
+                         c->step, c->eta, c->eta_snap, c->delta_snap,
+                         c->ctrl, total);
+      launch_polar(X, c->step, nullptr, 1.0, 1.0, 0.0, c->Xprop, n, d, r,
+                   c->ctrl, ST_TCG_STOP, s);
+      ctx_spmm(c, c->Xprop, c->W, ST_TCG_STOP, s);
+      hipLaunchKernelGGL(k_dots, dim3(gvec), dim3(256), 0, s,
+                         c->Xprop, c->W, c->Gt, c->ctrl, C_DOT0, C_DOT2,
+                         total, ST_TCG_STOP);
+      hipLaunchKernelGGL(k_ctrl_accept, dim3(1), dim3(64), 0, s, c->ctrl,
+                         accept_rho);
+      DPO_CHECK(hipMemcpyAsync(c->ctrl_host, c->ctrl,
+                               CTRL_SIZE * sizeof(double),
+                               hipMemcpyDeviceToHost, s));
+      DPO_CHECK(hipStreamSynchronize(s));
+    }
     int st = (int)c->ctrl_host[C_STATUS];
     if (st == ST_ACCEPTED) {
       DPO_CHECK(hipMemcpyAsync(X, c->Xprop, total * sizeof(double),
@@ -1325,7 +1403,6 @@ int dpo_rbcd_solve(void* h, double* X, double tol, double Delta0,
     }
     if (st == ST_NO_UPDATE) { status = st; break; }
     shrinks++;
-    hipLaunchKernelGGL(k_ctrl_shrink, dim3(1), dim3(64), 0, s, c->ctrl);
   }
 
   double f_init = c->ctrl_host[C_FX];
@@ -1355,6 +1432,14 @@ int dpo_rbcd_solve(void* h, double* X, double tol, double Delta0,
   return status;
 }
 
+int dpo_rbcd_solve(void* h, double* X, double tol, double Delta0,
+                   int max_shrink, double accept_rho,
+                   int compute_final_gn, double* stats_out, void* stream) {
+  return rbcd_solve_impl((DpoCtx*)h, X, nullptr, tol, Delta0, max_shrink,
+                         accept_rho, compute_final_gn, stats_out,
+                         (hipStream_t)stream);
+}
+
 // Per-round evaluation: out_dev (>=3 doubles, device) = [f, 0.5<X,G>, gn2]
 // using the ctx's problem pointers. No host sync.
 void dpo_eval_terms(void* h, const double* X, double* out_dev,
@@ -1382,6 +1467,7 @@ void dpo_eval_terms(void* h, const double* X, double* out_dev,
 void dpo_ctx_set_gdata(void* h, const double* E0, const long* local_pose,
                        const long* nbr_slot, const double* w, int ne) {
   DpoCtx* c = (DpoCtx*)h;
+  if (c->g_E0 != E0 || c->g_w != w) c->invalidate_graphs();
   c->g_E0 = E0; c->g_local_pose = local_pose; c->g_nbr_slot = nbr_slot;
   c->g_w = w; c->g_ne = ne;
 }
@@ -1391,17 +1477,42 @@ void dpo_ctx_set_gdata(void* h, const double* E0, const long* local_pose,
 int dpo_round_solve(void* h, double* X, const double* nbr, double tol,
                     double Delta0, int max_shrink, double accept_rho,
                     double* stats_out, void* stream) {
-  DpoCtx* c = (DpoCtx*)h;
-  ctx_assemble_g(c, nbr, (hipStream_t)stream);
-  return dpo_rbcd_solve(h, X, tol, Delta0, max_shrink, accept_rho, 0,
-                        stats_out, stream);
+  return rbcd_solve_impl((DpoCtx*)h, X, nbr, tol, Delta0, max_shrink,
+                         accept_rho, 0, stats_out, (hipStream_t)stream);
 }
 
 void dpo_round_eval(void* h, const double* X, const double* nbr,
                     double* out_dev, void* stream) {
   DpoCtx* c = (DpoCtx*)h;
-  ctx_assemble_g(c, nbr, (hipStream_t)stream);
-  dpo_eval_terms(h, X, out_dev, stream);
+  hipStream_t s = (hipStream_t)stream;
+  const void* key[4] = {X, nbr, out_dev, nullptr};
+  bool match = c->eval_graph && memcmp(key, c->eval_key, sizeof(key)) == 0;
+  if (!match) {
+    if (c->eval_graph) {
+      hipGraphExecDestroy(c->eval_graph);
+      c->eval_graph = nullptr;
+    }
+    hipGraph_t graph = nullptr;
+    hipError_t rc = hipStreamBeginCapture(
+        s, hipStreamCaptureModeThreadLocal);
+    if (rc == hipSuccess) {
+      ctx_assemble_g(c, nbr, s);
+      dpo_eval_terms(h, X, out_dev, stream);
+      rc = hipStreamEndCapture(s, &graph);
+    }
+    if (rc == hipSuccess && graph) {
+      rc = hipGraphInstantiate(&c->eval_graph, graph, nullptr, nullptr, 0);
+      hipGraphDestroy(graph);
+    }
+    if (rc != hipSuccess || !c->eval_graph) {
+      c->eval_graph = nullptr;
+      ctx_assemble_g(c, nbr, s);
+      dpo_eval_terms(h, X, out_dev, stream);
+      return;
+    }
+    memcpy(c->eval_key, key, sizeof(key));
+  }
+  DPO_CHECK(hipGraphLaunch(c->eval_graph, s));
 }
 
 }  // extern "C"
