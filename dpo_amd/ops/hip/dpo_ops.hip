@@ -1179,6 +1179,9 @@ struct DpoCtx {
   const void* solve_key[4] = {};
   hipGraphExec_t eval_graph = nullptr;
   const void* eval_key[4] = {};
+  // private stream used only for RECORDING captures (the legacy default
+  // stream cannot be captured); graphs replay on the caller's stream.
+  hipStream_t cap_stream = nullptr;
   void invalidate_graphs() {
     if (solve_graph) { hipGraphExecDestroy(solve_graph); solve_graph = nullptr; }
     if (eval_graph) { hipGraphExecDestroy(eval_graph); eval_graph = nullptr; }
@@ -1232,6 +1235,8 @@ void* dpo_ctx_create(int n, int d, int r, int max_inner) {
   DPO_CHECK(hipMalloc(&c->ctrl, CTRL_SIZE * sizeof(double)));
   DPO_CHECK(hipMalloc(&c->G_buf, vb));
   DPO_CHECK(hipHostMalloc(&c->ctrl_host, CTRL_SIZE * sizeof(double)));
+  DPO_CHECK(hipStreamCreateWithFlags(&c->cap_stream,
+                                     hipStreamNonBlocking));
   return c;
 }
 
@@ -1241,6 +1246,7 @@ void dpo_ctx_destroy(void* h) {
   hipFree(c->rvec); hipFree(c->z); hipFree(c->Hd); hipFree(c->step);
   hipFree(c->Xprop); hipFree(c->eta_snap); hipFree(c->delta_snap);
   c->invalidate_graphs();
+  if (c->cap_stream) hipStreamDestroy(c->cap_stream);
   hipFree(c->ctrl); hipFree(c->G_buf); hipHostFree(c->ctrl_host);
   delete c;
 }
@@ -1346,18 +1352,27 @@ static int rbcd_solve_impl(DpoCtx* c, double* X, const double* nbr,
   if (!key_match) {
     c->invalidate_graphs();
     hipGraph_t graph = nullptr;
+    hipStream_t cs_ = c->cap_stream;
     hipError_t rc = hipStreamBeginCapture(
-        s, hipStreamCaptureModeThreadLocal);
+        cs_, hipStreamCaptureModeThreadLocal);
     if (rc == hipSuccess) {
-      enqueue_solve_body(c, X, nbr, tol, Delta0, accept_rho, s);
-      rc = hipStreamEndCapture(s, &graph);
+      enqueue_solve_body(c, X, nbr, tol, Delta0, accept_rho, cs_);
+      rc = hipStreamEndCapture(cs_, &graph);
+    }
+    hipStreamCaptureStatus capst = hipStreamCaptureStatusNone;
+    hipStreamIsCapturing(cs_, &capst);
+    if (capst != hipStreamCaptureStatusNone) {
+      hipGraph_t dead = nullptr;
+      hipStreamEndCapture(cs_, &dead);
+      if (dead) hipGraphDestroy(dead);
     }
     if (rc == hipSuccess && graph) {
       rc = hipGraphInstantiate(&c->solve_graph, graph, nullptr, nullptr, 0);
       hipGraphDestroy(graph);
     }
+    hipGetLastError();  // clear sticky capture-related error state
     if (rc != hipSuccess || !c->solve_graph) {
-      // capture unavailable: run eagerly
+      // capture unavailable: run eagerly on the caller's stream
       c->solve_graph = nullptr;
       enqueue_solve_body(c, X, nbr, tol, Delta0, accept_rho, s);
       DPO_CHECK(hipStreamSynchronize(s));
@@ -1493,17 +1508,26 @@ void dpo_round_eval(void* h, const double* X, const double* nbr,
       c->eval_graph = nullptr;
     }
     hipGraph_t graph = nullptr;
+    hipStream_t cs_ = c->cap_stream;
     hipError_t rc = hipStreamBeginCapture(
-        s, hipStreamCaptureModeThreadLocal);
+        cs_, hipStreamCaptureModeThreadLocal);
     if (rc == hipSuccess) {
-      ctx_assemble_g(c, nbr, s);
-      dpo_eval_terms(h, X, out_dev, stream);
-      rc = hipStreamEndCapture(s, &graph);
+      ctx_assemble_g(c, nbr, cs_);
+      dpo_eval_terms(h, X, out_dev, (void*)cs_);
+      rc = hipStreamEndCapture(cs_, &graph);
+    }
+    hipStreamCaptureStatus capst = hipStreamCaptureStatusNone;
+    hipStreamIsCapturing(cs_, &capst);
+    if (capst != hipStreamCaptureStatusNone) {
+      hipGraph_t dead = nullptr;
+      hipStreamEndCapture(cs_, &dead);
+      if (dead) hipGraphDestroy(dead);
     }
     if (rc == hipSuccess && graph) {
       rc = hipGraphInstantiate(&c->eval_graph, graph, nullptr, nullptr, 0);
       hipGraphDestroy(graph);
     }
+    hipGetLastError();  // clear sticky capture-related error state
     if (rc != hipSuccess || !c->eval_graph) {
       c->eval_graph = nullptr;
       ctx_assemble_g(c, nbr, s);
