@@ -129,7 +129,8 @@ class PGOAgent:
                        shared_loop_closures: Sequence[RelativeSEMeasurement],
                        T_init: Optional[np.ndarray] = None) -> None:
         assert self.state == PGOAgentState.WAIT_FOR_DATA
-        if len(odometry) == 0:
+        if (len(odometry) == 0 and len(private_loop_closures) == 0
+                and len(shared_loop_closures) == 0):
             return
         for m in odometry:
             assert m.r1 == self.id and m.r2 == self.id and m.p1 + 1 == m.p2
@@ -236,7 +237,14 @@ class PGOAgent:
 
     def _local_initialization(self) -> None:
         meas = self.odometry + self.private_lc
-        if self.params.robust_cost_type == RobustCostType.L2:
+        if not meas:
+            # isolated / boundary-only agent: identity local init
+            dh = self.dh
+            T = np.zeros((self.d, self.n * dh))
+            for i in range(self.n):
+                T[:, i * dh:i * dh + self.d] = np.eye(self.d)
+            self.T_local_init = T
+        elif self.params.robust_cost_type == RobustCostType.L2:
             self.T_local_init = chordal_initialization(self.d, self.n, meas)
         else:
             # Robust mode: don't trust loop closures; odometry init
