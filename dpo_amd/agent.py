@@ -219,6 +219,11 @@ class PGOAgent:
         """Build the linear term from cached neighbor poses. Returns False
         (skip update) when any required pose is missing
         (PGOAgent.cpp:806-814)."""
+        if getattr(self, "_soa", False):
+            raise NotImplementedError(
+                "SoA agents (set_pose_graph_arrays) run through the packed "
+                "GPU path; the PoseDict exchange path needs object-mode "
+                "set_pose_graph")
         n_slots = len(self._nbr_slot_order)
         buf = np.zeros((n_slots, self.dh, self.r))
         for k, pid in enumerate(self._nbr_slot_order):
