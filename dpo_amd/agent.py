@@ -929,6 +929,21 @@ class PGOAgent:
         """Device 3-vector [f, 0.5<X,G>, ||rgrad||^2] with fresh G."""
         return self._dev_solver.round_eval(self.X, self._nbr_buffer)
 
+    def _packed_solve_async(self, accel: bool) -> None:
+        if accel:
+            self.X.copy_(self.Y)
+        nbr = self._nbr_buffer_aux if accel else self._nbr_buffer
+        self._dev_solver.round_solve_async(self.X, nbr)
+
+    def _packed_solve_finish(self) -> None:
+        self._dev_solver.round_solve_finish(self.X)
+
+    def _packed_eval_async(self):
+        return self._dev_solver.round_eval_async(self.X, self._nbr_buffer)
+
+    def _packed_eval_join(self):
+        return self._dev_solver.eval_join(self.X)
+
     # --- packed robust (GNC) support ----------------------------------
     def _packed_gnc_setup(self, dev) -> None:
         """Device tensors for the GNC weight-update kernel. SoA mode."""

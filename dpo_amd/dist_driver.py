@@ -561,20 +561,27 @@ class DistributedRBCDDriver:
                 aux_flats = self.comm.all_gather_flat(
                     self._packed_pack(use_aux=True), sizes)
                 self._packed_scatter(aux_flats, aux=True)
+            # concurrent active agents overlap on per-agent HIP streams
             for rb, a in self.local_agents.items():
                 if rb in active:
-                    a._packed_solve(accel)
+                    a._packed_solve_async(accel)
                 elif accel:
                     a.X.copy_(a.Y)
+            for rb in active:
+                if rb in self.local_agents:
+                    self.local_agents[rb]._packed_solve_finish()
             if accel:
                 for a in self.local_agents.values():
                     a._packed_nesterov_post(it)
             flats = self.comm.all_gather_flat(self._packed_pack(), sizes)
             self._packed_scatter(flats)
-            # evaluation (fresh neighbor data)
+            # evaluation (fresh neighbor data); agents fan out on their
+            # own streams and join back before the packed reduce
             evalmat.zero_()
             for rb, a in self.local_agents.items():
-                evalmat[rb] = a._packed_eval()
+                a._packed_eval_async()
+            for rb, a in self.local_agents.items():
+                evalmat[rb] = a._packed_eval_join()
             self.comm.all_reduce_sum_(evalmat)
             ev = evalmat.cpu().numpy()          # the round's one host sync
             cost = float((ev[:, 0] - ev[:, 1]).sum())

@@ -1182,6 +1182,16 @@ struct DpoCtx {
   // private stream used only for RECORDING captures (the legacy default
   // stream cannot be captured); graphs replay on the caller's stream.
   hipStream_t cap_stream = nullptr;
+  // per-agent execution stream + events: lets concurrent agents' solves
+  // and the per-round evaluations overlap on the GPU (colored schedule,
+  // 8-agent eval fan-out) while staying ordered against the caller's
+  // (torch) stream via events.
+  hipStream_t exec_stream = nullptr;
+  hipEvent_t start_event = nullptr;
+  hipEvent_t done_event = nullptr;
+  // state of an in-flight async solve
+  double pend_tol = 0, pend_Delta0 = 0, pend_rho = 0;
+  double* pend_X = nullptr;
   void invalidate_graphs() {
     if (solve_graph) { hipGraphExecDestroy(solve_graph); solve_graph = nullptr; }
     if (eval_graph) { hipGraphExecDestroy(eval_graph); eval_graph = nullptr; }
@@ -1237,6 +1247,12 @@ void* dpo_ctx_create(int n, int d, int r, int max_inner) {
   DPO_CHECK(hipHostMalloc(&c->ctrl_host, CTRL_SIZE * sizeof(double)));
   DPO_CHECK(hipStreamCreateWithFlags(&c->cap_stream,
                                      hipStreamNonBlocking));
+  DPO_CHECK(hipStreamCreateWithFlags(&c->exec_stream,
+                                     hipStreamNonBlocking));
+  DPO_CHECK(hipEventCreateWithFlags(&c->start_event,
+                                    hipEventDisableTiming));
+  DPO_CHECK(hipEventCreateWithFlags(&c->done_event,
+                                    hipEventDisableTiming));
   return c;
 }
 
@@ -1247,6 +1263,9 @@ void dpo_ctx_destroy(void* h) {
   hipFree(c->Xprop); hipFree(c->eta_snap); hipFree(c->delta_snap);
   c->invalidate_graphs();
   if (c->cap_stream) hipStreamDestroy(c->cap_stream);
+  if (c->exec_stream) hipStreamDestroy(c->exec_stream);
+  if (c->start_event) hipEventDestroy(c->start_event);
+  if (c->done_event) hipEventDestroy(c->done_event);
   hipFree(c->ctrl); hipFree(c->G_buf); hipHostFree(c->ctrl_host);
   delete c;
 }
@@ -1337,14 +1356,13 @@ static void enqueue_solve_body(DpoCtx* c, double* X, const double* nbr,
 // [status, f_init, gn_init, f_opt, gn_opt, rho, shrink_count, iters]
 // nbr != null: assemble G from the packed neighbor buffer first; the
 // whole pre-sync sequence is then served from a cached hipGraph.
-static int rbcd_solve_impl(DpoCtx* c, double* X, const double* nbr,
-                           double tol, double Delta0, int max_shrink,
-                           double accept_rho, int compute_final_gn,
-                           double* stats_out, hipStream_t s) {
-  const int n = c->n, d = c->d, r = c->r;
-  const long total = c->total;
-  const int gvec = (int)((total + 255) / 256);
-
+// Ensure the solve graph for (X, nbr, tol, Delta0) is cached; run the
+// pre-sync sequence (via graph replay or eagerly) on stream s WITHOUT
+// the final synchronize. Returns true when the sequence was enqueued
+// asynchronously (graph path) — eager fallback syncs internally.
+static bool solve_presync(DpoCtx* c, double* X, const double* nbr,
+                          double tol, double Delta0, double accept_rho,
+                          hipStream_t s) {
   const void* key[4] = {X, nbr, (const void*)(intptr_t)(tol * 1e9),
                         (const void*)(intptr_t)Delta0};
   bool key_match = c->solve_graph && memcmp(key, c->solve_key,
@@ -1376,16 +1394,23 @@ static int rbcd_solve_impl(DpoCtx* c, double* X, const double* nbr,
       c->solve_graph = nullptr;
       enqueue_solve_body(c, X, nbr, tol, Delta0, accept_rho, s);
       DPO_CHECK(hipStreamSynchronize(s));
-    } else {
-      memcpy(c->solve_key, key, sizeof(key));
+      return false;
     }
+    memcpy(c->solve_key, key, sizeof(key));
   }
-  if (c->solve_graph) {
-    DPO_CHECK(hipGraphLaunch(c->solve_graph, s));
-    DPO_CHECK(hipStreamSynchronize(s));
-  }
+  DPO_CHECK(hipGraphLaunch(c->solve_graph, s));
+  return true;
+}
 
-  // shrink loop (rare: only after a rejected first attempt)
+// Post-sync part of the solve: acceptance decision + (rare) shrink loop.
+// Assumes the pre-sync sequence has completed on stream s and ctrl_host
+// holds the control block.
+static int solve_postsync(DpoCtx* c, double* X, double accept_rho,
+                          int max_shrink, int compute_final_gn,
+                          double* stats_out, hipStream_t s) {
+  const int n = c->n, d = c->d, r = c->r;
+  const long total = c->total;
+  const int gvec = (int)((total + 255) / 256);
   int status = ST_GIVE_UP;
   int shrinks = 0;
   for (int attempt = 0; attempt <= max_shrink; ++attempt) {
@@ -1447,12 +1472,69 @@ static int rbcd_solve_impl(DpoCtx* c, double* X, const double* nbr,
   return status;
 }
 
+static int rbcd_solve_impl(DpoCtx* c, double* X, const double* nbr,
+                           double tol, double Delta0, int max_shrink,
+                           double accept_rho, int compute_final_gn,
+                           double* stats_out, hipStream_t s) {
+  bool async = solve_presync(c, X, nbr, tol, Delta0, accept_rho, s);
+  if (async) DPO_CHECK(hipStreamSynchronize(s));
+  return solve_postsync(c, X, accept_rho, max_shrink, compute_final_gn,
+                        stats_out, s);
+}
+
 int dpo_rbcd_solve(void* h, double* X, double tol, double Delta0,
                    int max_shrink, double accept_rho,
                    int compute_final_gn, double* stats_out, void* stream) {
   return rbcd_solve_impl((DpoCtx*)h, X, nullptr, tol, Delta0, max_shrink,
                          accept_rho, compute_final_gn, stats_out,
                          (hipStream_t)stream);
+}
+
+void dpo_round_eval(void* h, const double* X, const double* nbr,
+                    double* out_dev, void* stream);  // fwd decl
+
+// --- async (multi-stream) round entry points ------------------------
+// Launch the solve's pre-sync sequence on the ctx's private execution
+// stream, ordered after the caller's stream. Finish with
+// dpo_round_solve_finish. Concurrent agents overlap on the GPU.
+void dpo_round_solve_async(void* h, double* X, const double* nbr,
+                           double tol, double Delta0, double accept_rho,
+                           void* join_stream) {
+  DpoCtx* c = (DpoCtx*)h;
+  hipStream_t js = (hipStream_t)join_stream;
+  DPO_CHECK(hipEventRecord(c->start_event, js));
+  DPO_CHECK(hipStreamWaitEvent(c->exec_stream, c->start_event, 0));
+  solve_presync(c, X, nbr, tol, Delta0, accept_rho, c->exec_stream);
+  c->pend_X = X;
+  c->pend_tol = tol;
+  c->pend_Delta0 = Delta0;
+  c->pend_rho = accept_rho;
+}
+
+int dpo_round_solve_finish(void* h, int max_shrink, double* stats_out,
+                           void* join_stream) {
+  DpoCtx* c = (DpoCtx*)h;
+  DPO_CHECK(hipStreamSynchronize(c->exec_stream));
+  int st = solve_postsync(c, c->pend_X, c->pend_rho, max_shrink, 0,
+                          stats_out, c->exec_stream);
+  DPO_CHECK(hipEventRecord(c->done_event, c->exec_stream));
+  DPO_CHECK(hipStreamWaitEvent((hipStream_t)join_stream, c->done_event, 0));
+  return st;
+}
+
+void dpo_round_eval_async(void* h, const double* X, const double* nbr,
+                          double* out_dev, void* join_stream) {
+  DpoCtx* c = (DpoCtx*)h;
+  hipStream_t js = (hipStream_t)join_stream;
+  DPO_CHECK(hipEventRecord(c->start_event, js));
+  DPO_CHECK(hipStreamWaitEvent(c->exec_stream, c->start_event, 0));
+  dpo_round_eval(h, X, nbr, out_dev, (void*)c->exec_stream);
+  DPO_CHECK(hipEventRecord(c->done_event, c->exec_stream));
+}
+
+void dpo_eval_join(void* h, void* join_stream) {
+  DpoCtx* c = (DpoCtx*)h;
+  DPO_CHECK(hipStreamWaitEvent((hipStream_t)join_stream, c->done_event, 0));
 }
 
 // Per-round evaluation: out_dev (>=3 doubles, device) = [f, 0.5<X,G>, gn2]

@@ -61,6 +61,12 @@ _lib.dpo_round_solve.restype = _i
 _lib.dpo_round_solve.argtypes = [_c, _c, _c, _d, _d, _i, _d,
                                  ctypes.POINTER(ctypes.c_double), _c]
 _lib.dpo_round_eval.argtypes = [_c, _c, _c, _c, _c]
+_lib.dpo_round_solve_async.argtypes = [_c, _c, _c, _d, _d, _d, _c]
+_lib.dpo_round_solve_finish.restype = _i
+_lib.dpo_round_solve_finish.argtypes = [
+    _c, _i, ctypes.POINTER(ctypes.c_double), _c]
+_lib.dpo_round_eval_async.argtypes = [_c, _c, _c, _c, _c]
+_lib.dpo_eval_join.argtypes = [_c, _c]
 _lib.dpo_gnc_weights.argtypes = [_c, _c, _c, _c, _c, _c, _c, _c, _c, _c,
                                  _c, _c, _c, _i, _i, _i, _d, _d, _c]
 
@@ -289,6 +295,25 @@ class DeviceSolver:
     def round_eval(self, X: Tensor, nbr: Tensor) -> Tensor:
         _lib.dpo_round_eval(self.handle, _p(X), _p(nbr),
                             _p(self._eval_out), _stream(X))
+        return self._eval_out
+
+    # --- async multi-stream variants (overlap concurrent agents) -----
+    def round_solve_async(self, X: Tensor, nbr: Tensor, tol: float = 1e-2,
+                          Delta0: float = 100.0) -> None:
+        _lib.dpo_round_solve_async(self.handle, _p(X), _p(nbr), tol,
+                                   Delta0, 0.1, _stream(X))
+
+    def round_solve_finish(self, X: Tensor) -> int:
+        return _lib.dpo_round_solve_finish(self.handle, 10, self._stats,
+                                           _stream(X))
+
+    def round_eval_async(self, X: Tensor, nbr: Tensor) -> Tensor:
+        _lib.dpo_round_eval_async(self.handle, _p(X), _p(nbr),
+                                  _p(self._eval_out), _stream(X))
+        return self._eval_out
+
+    def eval_join(self, X: Tensor) -> Tensor:
+        _lib.dpo_eval_join(self.handle, _stream(X))
         return self._eval_out
 
 
