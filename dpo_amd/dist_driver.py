@@ -561,15 +561,28 @@ class DistributedRBCDDriver:
                 aux_flats = self.comm.all_gather_flat(
                     self._packed_pack(use_aux=True), sizes)
                 self._packed_scatter(aux_flats, aux=True)
-            # concurrent active agents overlap on per-agent HIP streams
+            # NOTE: the async multi-stream SOLVE path (DPO_ASYNC_SOLVE=1)
+            # is experimental: on ROCm 7.2 a cached hipGraph exec replayed
+            # on a per-agent stream intermittently degrades after ~40
+            # replays (garbage control-block reads; inputs verified
+            # finite, a freshly captured graph on the same state works).
+            # Until that driver-level interaction is understood the solve
+            # replays on the main stream; the EVAL fan-out (verified
+            # bitwise-identical) does overlap on per-agent streams.
+            import os as _os
+            _sync_solve = _os.environ.get("DPO_ASYNC_SOLVE", "0") != "1"
             for rb, a in self.local_agents.items():
                 if rb in active:
-                    a._packed_solve_async(accel)
+                    if _sync_solve:
+                        a._packed_solve(accel)
+                    else:
+                        a._packed_solve_async(accel)
                 elif accel:
                     a.X.copy_(a.Y)
-            for rb in active:
-                if rb in self.local_agents:
-                    self.local_agents[rb]._packed_solve_finish()
+            if not _sync_solve:
+                for rb in active:
+                    if rb in self.local_agents:
+                        self.local_agents[rb]._packed_solve_finish()
             if accel:
                 for a in self.local_agents.values():
                     a._packed_nesterov_post(it)
@@ -578,10 +591,18 @@ class DistributedRBCDDriver:
             # evaluation (fresh neighbor data); agents fan out on their
             # own streams and join back before the packed reduce
             evalmat.zero_()
-            for rb, a in self.local_agents.items():
-                a._packed_eval_async()
-            for rb, a in self.local_agents.items():
-                evalmat[rb] = a._packed_eval_join()
+            # eval fan-out on per-agent streams pays when several solves
+            # precede it (colored schedule); otherwise the event overhead
+            # outweighs the overlap (measured on MI355X).
+            _sync_eval_default = "0" if self.selection == "colored" else "1"
+            if _os.environ.get("DPO_SYNC_EVAL", _sync_eval_default) == "1":
+                for rb, a in self.local_agents.items():
+                    evalmat[rb] = a._packed_eval()
+            else:
+                for rb, a in self.local_agents.items():
+                    a._packed_eval_async()
+                for rb, a in self.local_agents.items():
+                    evalmat[rb] = a._packed_eval_join()
             self.comm.all_reduce_sum_(evalmat)
             ev = evalmat.cpu().numpy()          # the round's one host sync
             cost = float((ev[:, 0] - ev[:, 1]).sum())
