@@ -925,9 +925,9 @@ class PGOAgent:
         nbr = self._nbr_buffer_aux if accel else self._nbr_buffer
         self._dev_solver.round_solve(self.X, nbr, tol=1e-2, Delta0=100.0)
 
-    def _packed_eval(self):
+    def _packed_eval(self, out=None):
         """Device 3-vector [f, 0.5<X,G>, ||rgrad||^2] with fresh G."""
-        return self._dev_solver.round_eval(self.X, self._nbr_buffer)
+        return self._dev_solver.round_eval(self.X, self._nbr_buffer, out)
 
     def _packed_solve_async(self, accel: bool) -> None:
         if accel:
@@ -938,8 +938,9 @@ class PGOAgent:
     def _packed_solve_finish(self) -> None:
         self._dev_solver.round_solve_finish(self.X)
 
-    def _packed_eval_async(self):
-        return self._dev_solver.round_eval_async(self.X, self._nbr_buffer)
+    def _packed_eval_async(self, out=None):
+        return self._dev_solver.round_eval_async(self.X, self._nbr_buffer,
+                                                 out)
 
     def _packed_eval_join(self):
         return self._dev_solver.eval_join(self.X)

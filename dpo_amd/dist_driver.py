@@ -417,17 +417,9 @@ class DistributedRBCDDriver:
         dev = torch.device(self.device)
         blk = self.dh * self.r
         pk = {}
-        # per-agent public index tensors & rank payload layout
-        pk["pub_idx_t"] = {}
-        offs = {}
-        off = 0
-        for rb in self.rank_agents[self.comm.rank]:
-            offs[rb] = off
-            off += len(self.pub_idx[rb]) * blk
         pk["rank_payload_len"] = [
             sum(len(self.pub_idx[rb]) * blk for rb in agents)
             for agents in self.rank_agents]
-        pk["local_off"] = offs
         # global offsets: agent rb's block inside its owner's payload
         glob_off = {}
         for rk, agents in enumerate(self.rank_agents):
@@ -436,52 +428,101 @@ class DistributedRBCDDriver:
                 glob_off[rb] = o
                 o += len(self.pub_idx[rb]) * blk
         pk["glob_off"] = glob_off
-        # per local agent: for each neighbor agent, (src positions within
-        # neighbor pub list, dst slots in local nbr buffer)
-        pk["scatter"] = {}
-        for rb, a in self.local_agents.items():
-            a._ensure_packed(dev)
-            pk["pub_idx_t"][rb] = torch.tensor(
-                self.pub_idx[rb], dtype=torch.int64, device=dev)
-            per_nbr = {}
-            pos_in_pub = {nb: {p: k for k, p in enumerate(self.pub_idx[nb])}
-                          for nb in a.get_neighbors()}
+
+        # ---- consolidated per-rank buffers --------------------------
+        # One X buffer and one neighbor buffer for all local agents:
+        # pack/scatter become single index ops, and each agent's tensors
+        # are stable contiguous views (the hipGraph keys stay valid).
+        my = self.rank_agents[self.comm.rank]
+        for rb in my:
+            self.local_agents[rb]._ensure_packed(dev)
+        pose_off = {}
+        off = 0
+        for rb in my:
+            pose_off[rb] = off
+            off += self.local_agents[rb].n
+        rank_X = torch.zeros(off * self.dh, self.r, dtype=torch.float64,
+                             device=dev)
+        for rb in my:
+            a = self.local_agents[rb]
+            o = pose_off[rb] * self.dh
+            rank_X[o:o + a.n * self.dh].copy_(a.X)
+            a.X = rank_X[o:o + a.n * self.dh]
+            if a.params.acceleration:
+                a.Y = a.X.clone()
+                a.V = a.X.clone()
+        pk["rank_X"] = rank_X
+        slot_off = {}
+        soff = 0
+        for rb in my:
+            slot_off[rb] = soff
+            soff += max(len(self.local_agents[rb]._nbr_slot_order), 1)
+        rank_nbr = torch.zeros(max(soff, 1), self.dh, self.r,
+                               dtype=torch.float64, device=dev)
+        rank_nbr_aux = torch.zeros_like(rank_nbr)
+        for rb in my:
+            a = self.local_agents[rb]
+            ns = max(len(a._nbr_slot_order), 1)
+            a._nbr_buffer = rank_nbr[slot_off[rb]:slot_off[rb] + ns]
+            a._nbr_buffer_aux = rank_nbr_aux[slot_off[rb]:slot_off[rb] + ns]
+        pk["rank_nbr"] = rank_nbr
+        pk["rank_nbr_aux"] = rank_nbr_aux
+
+        # single pack index: global pose-block rows of every public pose
+        pack_rows = []
+        for rb in my:
+            for p in self.pub_idx[rb]:
+                pack_rows.append(pose_off[rb] + p)
+        pk["pack_rows"] = torch.tensor(pack_rows, dtype=torch.int64,
+                                       device=dev)
+        # scatter: one (src positions -> nbr slots) pair per SOURCE rank
+        pub_payload_pos = {}  # (nb agent, pose) -> index in owner payload
+        for rb in range(self.num_robots):
+            base = glob_off[rb] // blk
+            for k, p in enumerate(self.pub_idx[rb]):
+                pub_payload_pos[(rb, p)] = base + k
+        per_src = {}
+        for rb in my:
+            a = self.local_agents[rb]
             for slot, (nb, p) in enumerate(a._nbr_slot_order):
-                per_nbr.setdefault(nb, ([], []))
-                per_nbr[nb][0].append(pos_in_pub[nb][p])
-                per_nbr[nb][1].append(slot)
-            pk["scatter"][rb] = {
-                nb: (torch.tensor(src, dtype=torch.int64, device=dev),
-                     torch.tensor(dst, dtype=torch.int64, device=dev))
-                for nb, (src, dst) in per_nbr.items()}
+                rk = self.owner[nb]
+                per_src.setdefault(rk, ([], []))
+                per_src[rk][0].append(pub_payload_pos[(nb, p)])
+                per_src[rk][1].append(slot_off[rb] + slot)
+        pk["scatter_rank"] = {
+            rk: (torch.tensor(src, dtype=torch.int64, device=dev),
+                 torch.tensor(dst, dtype=torch.int64, device=dev))
+            for rk, (src, dst) in per_src.items()}
         pk["dev"] = dev
         self._pk = pk
 
     def _packed_pack(self, use_aux: bool = False):
         import torch
         pk = self._pk
-        blk = self.dh * self.r
-        parts = []
-        for rb in self.rank_agents[self.comm.rank]:
-            a = self.local_agents[rb]
-            src = a.Y if (use_aux and a.Y is not None) else a.X
-            Xb = src.view(a.n, self.dh, self.r)
-            parts.append(Xb.index_select(0, pk["pub_idx_t"][rb]).reshape(-1))
-        if parts:
-            return torch.cat(parts)
-        return torch.zeros(0, dtype=torch.float64, device=pk["dev"])
+        if not self.rank_agents[self.comm.rank]:
+            return torch.zeros(0, dtype=torch.float64, device=pk["dev"])
+        if use_aux:
+            parts = []
+            for rb in self.rank_agents[self.comm.rank]:
+                a = self.local_agents[rb]
+                src = a.Y if a.Y is not None else a.X
+                n_pub = len(self.pub_idx[rb])
+                Xb = src.view(a.n, self.dh, self.r)
+                idx = torch.tensor(self.pub_idx[rb], dtype=torch.int64,
+                                   device=pk["dev"])
+                parts.append(Xb.index_select(0, idx).reshape(-1))
+            return torch.cat(parts) if parts else torch.zeros(
+                0, dtype=torch.float64, device=pk["dev"])
+        Xb = pk["rank_X"].view(-1, self.dh, self.r)
+        return Xb.index_select(0, pk["pack_rows"]).reshape(-1)
 
     def _packed_scatter(self, flats, aux: bool = False):
         pk = self._pk
         blk = self.dh * self.r
-        for rb, a in self.local_agents.items():
-            for nb, (src_idx, dst_slots) in pk["scatter"][rb].items():
-                payload = flats[self.owner[nb]]
-                o = pk["glob_off"][nb]
-                npub = len(self.pub_idx[nb])
-                blkv = payload[o:o + npub * blk].view(npub, self.dh, self.r)
-                buf = a._nbr_buffer_aux if aux else a._nbr_buffer
-                buf.index_copy_(0, dst_slots, blkv.index_select(0, src_idx))
+        target = pk["rank_nbr_aux"] if aux else pk["rank_nbr"]
+        for rk, (src_idx, dst_slots) in pk["scatter_rank"].items():
+            blkv = flats[rk].view(-1, self.dh, self.r)
+            target.index_copy_(0, dst_slots, blkv.index_select(0, src_idx))
 
     def _run_packed(self, max_iters, gradnorm_tol, trace_file, time_limit_s):
         import torch

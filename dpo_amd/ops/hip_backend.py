@@ -292,10 +292,12 @@ class DeviceSolver:
                                     Delta0, 10, 0.1, self._stats,
                                     _stream(X))
 
-    def round_eval(self, X: Tensor, nbr: Tensor) -> Tensor:
-        _lib.dpo_round_eval(self.handle, _p(X), _p(nbr),
-                            _p(self._eval_out), _stream(X))
-        return self._eval_out
+    def round_eval(self, X: Tensor, nbr: Tensor,
+                   out: Optional[Tensor] = None) -> Tensor:
+        dst = self._eval_out if out is None else out
+        _lib.dpo_round_eval(self.handle, _p(X), _p(nbr), _p(dst),
+                            _stream(X))
+        return dst
 
     # --- async multi-stream variants (overlap concurrent agents) -----
     def round_solve_async(self, X: Tensor, nbr: Tensor, tol: float = 1e-2,
@@ -307,10 +309,12 @@ class DeviceSolver:
         return _lib.dpo_round_solve_finish(self.handle, 10, self._stats,
                                            _stream(X))
 
-    def round_eval_async(self, X: Tensor, nbr: Tensor) -> Tensor:
-        _lib.dpo_round_eval_async(self.handle, _p(X), _p(nbr),
-                                  _p(self._eval_out), _stream(X))
-        return self._eval_out
+    def round_eval_async(self, X: Tensor, nbr: Tensor,
+                         out: Optional[Tensor] = None) -> Tensor:
+        dst = self._eval_out if out is None else out
+        _lib.dpo_round_eval_async(self.handle, _p(X), _p(nbr), _p(dst),
+                                  _stream(X))
+        return dst
 
     def eval_join(self, X: Tensor) -> Tensor:
         _lib.dpo_eval_join(self.handle, _stream(X))
