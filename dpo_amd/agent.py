@@ -926,8 +926,13 @@ class PGOAgent:
         self._dev_solver.round_solve(self.X, nbr, tol=1e-2, Delta0=100.0)
 
     def _packed_eval(self, out=None):
-        """Device 3-vector [f, 0.5<X,G>, ||rgrad||^2] with fresh G."""
-        return self._dev_solver.round_eval(self.X, self._nbr_buffer, out)
+        """Device 3-vector [f, 0.5<X,G>, ||rgrad||^2] with fresh G.
+        With an explicit out row the RAW enqueue is used so the whole
+        phase can be captured into one driver-level graph."""
+        if out is not None:
+            self._dev_solver.round_eval_raw(self.X, self._nbr_buffer, out)
+            return out
+        return self._dev_solver.round_eval(self.X, self._nbr_buffer)
 
     def _packed_solve_async(self, accel: bool) -> None:
         if accel:
@@ -1042,6 +1047,7 @@ class PGOAgent:
                       self._q_dev["edge_of"], self._all_weights_dev, self.dh)
         self.problem.refresh_preconditioner()
         self._dev_solver.bind_problem_static(self.problem)
+        self._packed_generation = getattr(self, "_packed_generation", 0) + 1
 
     def _packed_shared_weights(self) -> Tensor:
         return self._w_shared_dev

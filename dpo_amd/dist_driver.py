@@ -524,6 +524,40 @@ class DistributedRBCDDriver:
             blkv = flats[rk].view(-1, self.dh, self.r)
             target.index_copy_(0, dst_slots, blkv.index_select(0, src_idx))
 
+    def _packed_eval_phase(self, evalmat):
+        """Per-round evaluation of every local agent. The whole phase
+        (zeroing + every agent's G assembly + cost/gradient kernels,
+        writing directly into evalmat rows) is captured once into a
+        single hipGraph via torch.cuda.graph and replayed per round;
+        re-captured when an agent's problem pointers change (GNC Q
+        rebuild)."""
+        import os as _os
+        import torch
+        gen = sum(getattr(a, "_packed_generation", 0)
+                  for a in self.local_agents.values())
+        cache = getattr(self, "_eval_graph", None)
+        if _os.environ.get("DPO_NO_EVAL_GRAPH", "0") == "1":
+            evalmat.zero_()
+            for rb, a in self.local_agents.items():
+                a._packed_eval(out=evalmat[rb])
+            return
+        if cache is None or cache[1] != gen:
+            g = torch.cuda.CUDAGraph()
+            # warmup on a side stream (required before capture)
+            side = torch.cuda.Stream()
+            with torch.cuda.stream(side):
+                evalmat.zero_()
+                for rb, a in self.local_agents.items():
+                    a._packed_eval(out=evalmat[rb])
+            torch.cuda.current_stream().wait_stream(side)
+            with torch.cuda.graph(g):
+                evalmat.zero_()
+                for rb, a in self.local_agents.items():
+                    a._packed_eval(out=evalmat[rb])
+            self._eval_graph = (g, gen)
+            cache = self._eval_graph
+        cache[0].replay()
+
     def _run_packed(self, max_iters, gradnorm_tol, trace_file, time_limit_s):
         import torch
         from .types import OptAlgorithm
@@ -631,19 +665,7 @@ class DistributedRBCDDriver:
             self._packed_scatter(flats)
             # evaluation (fresh neighbor data); agents fan out on their
             # own streams and join back before the packed reduce
-            evalmat.zero_()
-            # eval fan-out on per-agent streams pays when several solves
-            # precede it (colored schedule); otherwise the event overhead
-            # outweighs the overlap (measured on MI355X).
-            _sync_eval_default = "0" if self.selection == "colored" else "1"
-            if _os.environ.get("DPO_SYNC_EVAL", _sync_eval_default) == "1":
-                for rb, a in self.local_agents.items():
-                    evalmat[rb] = a._packed_eval()
-            else:
-                for rb, a in self.local_agents.items():
-                    a._packed_eval_async()
-                for rb, a in self.local_agents.items():
-                    evalmat[rb] = a._packed_eval_join()
+            self._packed_eval_phase(evalmat)
             self.comm.all_reduce_sum_(evalmat)
             ev = evalmat.cpu().numpy()          # the round's one host sync
             cost = float((ev[:, 0] - ev[:, 1]).sum())

@@ -1578,6 +1578,15 @@ int dpo_round_solve(void* h, double* X, const double* nbr, double tol,
                          accept_rho, 0, stats_out, (hipStream_t)stream);
 }
 
+// Non-caching eval enqueue (for embedding in an externally captured
+// graph, e.g. one driver-level graph covering every agent's eval).
+void dpo_round_eval_raw(void* h, const double* X, const double* nbr,
+                        double* out_dev, void* stream) {
+  DpoCtx* c = (DpoCtx*)h;
+  ctx_assemble_g(c, nbr, (hipStream_t)stream);
+  dpo_eval_terms(h, X, out_dev, stream);
+}
+
 void dpo_round_eval(void* h, const double* X, const double* nbr,
                     double* out_dev, void* stream) {
   DpoCtx* c = (DpoCtx*)h;

@@ -61,6 +61,7 @@ _lib.dpo_round_solve.restype = _i
 _lib.dpo_round_solve.argtypes = [_c, _c, _c, _d, _d, _i, _d,
                                  ctypes.POINTER(ctypes.c_double), _c]
 _lib.dpo_round_eval.argtypes = [_c, _c, _c, _c, _c]
+_lib.dpo_round_eval_raw.argtypes = [_c, _c, _c, _c, _c]
 _lib.dpo_round_solve_async.argtypes = [_c, _c, _c, _d, _d, _d, _c]
 _lib.dpo_round_solve_finish.restype = _i
 _lib.dpo_round_solve_finish.argtypes = [
@@ -291,6 +292,10 @@ class DeviceSolver:
         return _lib.dpo_round_solve(self.handle, _p(X), _p(nbr), tol,
                                     Delta0, 10, 0.1, self._stats,
                                     _stream(X))
+
+    def round_eval_raw(self, X: Tensor, nbr: Tensor, out: Tensor) -> None:
+        _lib.dpo_round_eval_raw(self.handle, _p(X), _p(nbr), _p(out),
+                                _stream(X))
 
     def round_eval(self, X: Tensor, nbr: Tensor,
                    out: Optional[Tensor] = None) -> Tensor:
