@@ -541,6 +541,12 @@ class DistributedRBCDDriver:
             for rb, a in self.local_agents.items():
                 a._packed_eval(out=evalmat[rb])
             return
+        # ROCm 7.2: a long-lived graph exec intermittently degrades after
+        # a few hundred replays (stale node state; fresh capture of the
+        # identical sequence is correct). Re-capture periodically —
+        # amortized cost ~1%.
+        if cache is not None and cache[2] >= 128:
+            cache = None
         if cache is None or cache[1] != gen:
             g = torch.cuda.CUDAGraph()
             # warmup on a side stream (required before capture)
@@ -554,9 +560,10 @@ class DistributedRBCDDriver:
                 evalmat.zero_()
                 for rb, a in self.local_agents.items():
                     a._packed_eval(out=evalmat[rb])
-            self._eval_graph = (g, gen)
+            self._eval_graph = [g, gen, 0]
             cache = self._eval_graph
         cache[0].replay()
+        cache[2] += 1
 
     def _run_packed(self, max_iters, gradnorm_tol, trace_file, time_limit_s):
         import torch
