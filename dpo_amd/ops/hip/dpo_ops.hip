@@ -1177,8 +1177,13 @@ struct DpoCtx {
   // them (preconditioner refresh, Q rebuild keep the same buffers).
   hipGraphExec_t solve_graph = nullptr;
   const void* solve_key[4] = {};
+  int solve_replays = 0;
   hipGraphExec_t eval_graph = nullptr;
   const void* eval_key[4] = {};
+  int eval_replays = 0;
+  // ROCm 7.2: long-lived graph execs intermittently degrade after a few
+  // hundred replays; refresh them periodically (amortized ~1%).
+  static constexpr int kMaxReplays = 128;
   // private stream used only for RECORDING captures (the legacy default
   // stream cannot be captured); graphs replay on the caller's stream.
   hipStream_t cap_stream = nullptr;
@@ -1366,9 +1371,14 @@ static bool solve_presync(DpoCtx* c, double* X, const double* nbr,
   const void* key[4] = {X, nbr, (const void*)(intptr_t)(tol * 1e9),
                         (const void*)(intptr_t)Delta0};
   bool key_match = c->solve_graph && memcmp(key, c->solve_key,
-                                            sizeof(key)) == 0;
+                                            sizeof(key)) == 0
+                   && c->solve_replays < DpoCtx::kMaxReplays;
   if (!key_match) {
-    c->invalidate_graphs();
+    if (c->solve_graph) {
+      hipGraphExecDestroy(c->solve_graph);
+      c->solve_graph = nullptr;
+    }
+    c->solve_replays = 0;
     hipGraph_t graph = nullptr;
     hipStream_t cs_ = c->cap_stream;
     hipError_t rc = hipStreamBeginCapture(
@@ -1398,6 +1408,7 @@ static bool solve_presync(DpoCtx* c, double* X, const double* nbr,
     }
     memcpy(c->solve_key, key, sizeof(key));
   }
+  c->solve_replays++;
   DPO_CHECK(hipGraphLaunch(c->solve_graph, s));
   return true;
 }
@@ -1592,12 +1603,14 @@ void dpo_round_eval(void* h, const double* X, const double* nbr,
   DpoCtx* c = (DpoCtx*)h;
   hipStream_t s = (hipStream_t)stream;
   const void* key[4] = {X, nbr, out_dev, nullptr};
-  bool match = c->eval_graph && memcmp(key, c->eval_key, sizeof(key)) == 0;
+  bool match = c->eval_graph && memcmp(key, c->eval_key, sizeof(key)) == 0
+               && c->eval_replays < DpoCtx::kMaxReplays;
   if (!match) {
     if (c->eval_graph) {
       hipGraphExecDestroy(c->eval_graph);
       c->eval_graph = nullptr;
     }
+    c->eval_replays = 0;
     hipGraph_t graph = nullptr;
     hipStream_t cs_ = c->cap_stream;
     hipError_t rc = hipStreamBeginCapture(
@@ -1627,6 +1640,7 @@ void dpo_round_eval(void* h, const double* X, const double* nbr,
     }
     memcpy(c->eval_key, key, sizeof(key));
   }
+  c->eval_replays++;
   DPO_CHECK(hipGraphLaunch(c->eval_graph, s));
 }
 
