@@ -762,11 +762,18 @@ class PGOAgent:
         self.end_optimization_loop()
         if self.logger:
             meas = self.odometry + self.private_lc + self.shared_lc
-            self.logger.log_measurements(meas, "measurements.csv")
+            if meas:
+                self.logger.log_measurements(meas, "measurements.csv")
             T = self.get_trajectory_in_global_frame()
             if T is not None:
                 self.logger.log_trajectory(self.d, self.n, T,
                                            "trajectory_optimized.csv")
+            # pre-rounding lifted estimate (checkpoint; reference writes
+            # X.txt at PGOAgent.cpp:602) - resume via numpy load + set_x
+            if self.X is not None and self.logger.log_dir:
+                import os as _os
+                np.save(_os.path.join(self.logger.log_dir, "X.npy"),
+                        self.X.cpu().numpy().T)
         self.instance_number += 1
         self.iteration_number = 0
         self.num_poses_received = 0
