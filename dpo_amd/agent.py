@@ -918,14 +918,6 @@ class PGOAgent:
             self.alpha = 0.0
         self._packed_ready = True
 
-    def _packed_set_g(self, aux: bool) -> None:
-        if self.shared_lc:
-            buf = self._nbr_buffer_aux if aux else self._nbr_buffer
-            Gt = self._g_assembler.assemble(buf, self._w_shared_dev, self.r)
-            self.problem.set_g(Gt)
-        else:
-            self.problem.set_g(None)
-
     def _packed_solve(self, accel: bool) -> None:
         if accel:
             self.X.copy_(self.Y)

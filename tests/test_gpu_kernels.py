@@ -278,3 +278,43 @@ def test_soa_l2_gpu_matches_object_path():
     # init) — compare converged cost instead of traces
     assert abs(res_soa.final_cost - res_obj.final_cost) < 1e-2 * max(
         1.0, abs(res_obj.final_cost))
+
+
+def test_async_optimization_loop_gpu():
+    """Asynchronous per-agent optimization thread on the GPU path
+    (reference testOptimizationThread semantics)."""
+    import time
+    from dpo_amd.agent import PGOAgent
+    from dpo_amd.synthetic import triangle_graph
+    from dpo_amd.types import PGOAgentParams
+    import numpy as np
+    meas, n, T_truth = triangle_graph()
+    odo = [m for m in meas if m.p1 + 1 == m.p2]
+    lc = [m for m in meas if m.p1 + 1 != m.p2]
+    a = PGOAgent(0, PGOAgentParams(d=3, r=5, device=DEV))
+    a.set_pose_graph(odo, lc, [])
+    a.start_optimization_loop(100.0)
+    time.sleep(0.5)
+    a.end_optimization_loop()
+    T = a.get_trajectory_in_local_frame()
+    assert np.abs(T - T_truth).max() < 1e-4
+
+
+def test_single_robot_batch_gpu_matches_cpu():
+    """Full-batch RTR (r = d) on GPU vs CPU (reference
+    SingleRobotExample path)."""
+    from dpo_amd.agent import PGOAgent
+    from dpo_amd.synthetic import grid3d
+    from dpo_amd.types import PGOAgentParams
+    meas, n = grid3d(side=3, seed=1)
+    odo = [m for m in meas if m.p1 + 1 == m.p2]
+    lc = [m for m in meas if m.p1 + 1 != m.p2]
+    res = {}
+    for dev in ("cpu", DEV):
+        a = PGOAgent(0, PGOAgentParams(d=3, r=3, device=dev))
+        a.set_pose_graph(odo, lc, [])
+        a.local_pose_graph_optimization()
+        res[dev] = a.last_opt_result
+    assert abs(res[DEV].f_opt - res["cpu"].f_opt) < 1e-4 * max(
+        1.0, abs(res["cpu"].f_opt))
+    assert res[DEV].grad_norm_opt < 1e-1
