@@ -91,6 +91,130 @@ __device__ __forceinline__ bool guarded_off(const double* ctrl, int guard) {
   return guard >= 0 && ctrl != nullptr && ctrl[C_STATUS] != (double)guard;
 }
 
+#define C_TICKET C_DOT3  // fan-in arrival counter (scratch slot)
+
+// L2-coherent (L1-bypassing) ctrl accesses for the fused control tails:
+// other workgroups' dot contributions arrive as device-scope atomics, so
+// the last-arriving block must read them past its own L1.
+__device__ __forceinline__ double ctrl_load(const double* p) {
+  return __hip_atomic_load(p, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+}
+__device__ __forceinline__ void ctrl_store(double* p, double v) {
+  __hip_atomic_store(p, v, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+}
+
+// Returns true for exactly one block — the last to arrive — after every
+// block's prior vector-memory ops (incl. the dot atomics) have retired.
+// Split-K fan-in (guide G16): per-block vmcnt drain, then a device-scope
+// ticket; the winner resets the ticket for the next fused kernel.
+__device__ __forceinline__ bool fanin_last_block(double* ctrl) {
+  __syncthreads();
+  __shared__ int is_last;
+  if (threadIdx.x == 0) {
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    double old = atomicAdd(ctrl + C_TICKET, 1.0);
+    is_last = ((int)old == (int)gridDim.x - 1) ? 1 : 0;
+    if (is_last) ctrl_store(ctrl + C_TICKET, 0.0);
+  }
+  __syncthreads();
+  return is_last != 0;
+}
+
+// Control tails (single thread of the last block). Mirror the k_ctrl_*
+// kernels; dot slots are read L2-coherently, results stored L2-visibly
+// (the next kernel's dispatch acquire makes them visible to all CUs).
+enum CtrlFuse { CF_NONE = 0, CF_Z0 = 1, CF_ALPHA = 2, CF_RR = 3,
+                CF_BETA = 4, CF_COMBINE = 5 };
+
+__device__ void ctrl_tail_z0(double* ctrl) {
+  if (ctrl[C_STATUS] != (double)ST_RUN) return;
+  const double zr = ctrl_load(ctrl + C_DOT0);
+  ctrl_store(ctrl + C_ZR, zr);
+  ctrl_store(ctrl + C_DPD, zr);
+  ctrl_store(ctrl + C_DOT0, 0.0);
+}
+
+__device__ void ctrl_tail_alpha(double* ctrl) {
+  if (ctrl[C_STATUS] != (double)ST_RUN) return;
+  const int j = (int)ctrl[C_ITER];
+  const double d_Hd = ctrl_load(ctrl + C_DOT0);
+  ctrl_store(ctrl + C_DOT0, 0.0);
+  const double z_r = ctrl[C_ZR];
+  const double e_Pe = ctrl[C_EPE], e_Pd = ctrl[C_EPD], d_Pd = ctrl[C_DPD];
+  const double radius = ctrl[C_RADIUS];
+  ctrl_store(ctrl + H_ZR(j), z_r);
+  ctrl_store(ctrl + H_DHD(j), d_Hd);
+  ctrl_store(ctrl + H_EPE(j), e_Pe);
+  ctrl_store(ctrl + H_EPD(j), e_Pd);
+  ctrl_store(ctrl + H_DPD(j), d_Pd);
+  const double alpha = z_r / d_Hd;
+  const double e_Pe_new = e_Pe + 2.0 * alpha * e_Pd + alpha * alpha * d_Pd;
+  if (d_Hd <= 0.0 || e_Pe_new >= radius * radius) {
+    const double disc = e_Pd * e_Pd + d_Pd * (radius * radius - e_Pe);
+    const double tau = (d_Pd > 0.0)
+        ? (-e_Pd + sqrt(fmax(disc, 0.0))) / d_Pd : 0.0;
+    ctrl_store(ctrl + C_COEF, tau);
+    ctrl_store(ctrl + C_STOP_PENDING, 1.0);
+  } else {
+    ctrl_store(ctrl + C_COEF, alpha);
+    ctrl_store(ctrl + C_EPE, e_Pe_new);
+    ctrl_store(ctrl + C_STOP_PENDING, 0.0);
+  }
+}
+
+__device__ void ctrl_tail_rr(double* ctrl) {
+  if (ctrl[C_STATUS] != (double)ST_RUN) return;
+  const int j = (int)ctrl[C_ITER];
+  if (ctrl[C_STOP_PENDING] != 0.0) {
+    ctrl_store(ctrl + C_STATUS, (double)ST_TCG_STOP);
+    ctrl_store(ctrl + C_J, (double)j);
+    ctrl_store(ctrl + C_HLEN, (double)(j + 1));
+    ctrl_store(ctrl + C_USE_CURRENT, 0.0);
+    return;
+  }
+  const double rr = ctrl_load(ctrl + C_DOT1);
+  ctrl_store(ctrl + C_DOT1, 0.0);
+  ctrl_store(ctrl + C_RR, rr);
+  if (sqrt(rr) <= ctrl[C_BOUND]) {
+    ctrl_store(ctrl + C_STATUS, (double)ST_TCG_STOP);
+    ctrl_store(ctrl + C_J, (double)(j + 1));
+    ctrl_store(ctrl + C_HLEN, (double)(j + 1));
+    ctrl_store(ctrl + C_USE_CURRENT, 1.0);
+  }
+}
+
+__device__ void ctrl_tail_beta(double* ctrl) {
+  if (ctrl[C_STATUS] != (double)ST_RUN) return;
+  const double z_r_new = ctrl_load(ctrl + C_DOT0);
+  ctrl_store(ctrl + C_DOT0, 0.0);
+  const double z_r = ctrl[C_ZR];
+  const double alpha = ctrl[C_COEF];
+  const double beta = z_r_new / z_r;
+  ctrl_store(ctrl + C_BETA, beta);
+  ctrl_store(ctrl + C_EPD, beta * (ctrl[C_EPD] + alpha * ctrl[C_DPD]));
+  ctrl_store(ctrl + C_DPD, z_r_new + beta * beta * ctrl[C_DPD]);
+  ctrl_store(ctrl + C_ZR, z_r_new);
+  ctrl_store(ctrl + C_ITER, ctrl[C_ITER] + 1.0);
+}
+
+__device__ void ctrl_tail_combine(double* ctrl, double* out) {
+  out[0] = 0.5 * (ctrl_load(ctrl + C_DOT0) + ctrl_load(ctrl + C_DOT2));
+  out[1] = 0.5 * ctrl_load(ctrl + C_DOT2);
+  out[2] = ctrl_load(ctrl + C_DOT1);
+}
+
+__device__ __forceinline__ void run_ctrl_tail(int cf, double* ctrl,
+                                              double* out) {
+  switch (cf) {
+    case CF_Z0: ctrl_tail_z0(ctrl); break;
+    case CF_ALPHA: ctrl_tail_alpha(ctrl); break;
+    case CF_RR: ctrl_tail_rr(ctrl); break;
+    case CF_BETA: ctrl_tail_beta(ctrl); break;
+    case CF_COMBINE: ctrl_tail_combine(ctrl, out); break;
+    default: break;
+  }
+}
+
 // wave-level + block-level reduction, then one atomicAdd per block
 __device__ __forceinline__ void block_reduce_atomic(double v, double* dst) {
   __shared__ double sh[16];  // up to 1024/64 waves
@@ -107,6 +231,9 @@ __device__ __forceinline__ void block_reduce_atomic(double v, double* dst) {
       s += __shfl_down(s, off, 64);
     if (lane == 0 && s != 0.0) atomicAdd(dst, s);
   }
+  // the scratch is reused by back-to-back reductions in one kernel:
+  // wave 0 must finish reading before anyone writes the next round
+  __syncthreads();
 }
 
 // ---------------------------------------------------------------------
@@ -186,7 +313,7 @@ __global__ void k_bsr_spmm_gen(const int* __restrict__ row_ptr,
 #define DPO_MAX_R 8
 #define DPO_MAX_DH 4
 
-template <int NEG, int D, int R>
+template <int NEG, int D, int R, int CF = CF_NONE>
 __global__ void k_proj_dots(const double* __restrict__ X,
                             const double* __restrict__ V,
                             const double* __restrict__ G,
@@ -267,6 +394,10 @@ __global__ void k_proj_dots(const double* __restrict__ X,
   }
   if (dot_slot >= 0) block_reduce_atomic(dot_pw, ctrl + dot_slot);
   if (dot_slot2 >= 0) block_reduce_atomic(dot_vx, ctrl + dot_slot2);
+  if (CF != CF_NONE) {
+    if (fanin_last_block(ctrl) && threadIdx.x == 0)
+      run_ctrl_tail(CF, ctrl, nullptr);
+  }
 }
 
 // ---------------------------------------------------------------------
@@ -496,6 +627,7 @@ __global__ void k_precond_jacobi(const double* __restrict__ L,
 //   snapshot: eta_snap[j] = eta, delta_snap[j] = delta  (pre-update)
 //   eta += coef * delta;  if (!stop_pending) { r += alpha*Hd; rr+=r^2 }
 // ---------------------------------------------------------------------
+template <int CF = CF_NONE>
 __global__ void k_tcg_update(double* __restrict__ eta,
                              double* __restrict__ rvec,
                              const double* __restrict__ delta,
@@ -523,6 +655,10 @@ __global__ void k_tcg_update(double* __restrict__ eta,
     }
   }
   if (!stop_pending) block_reduce_atomic(rr, ctrl + C_DOT1);
+  if (CF != CF_NONE) {
+    if (fanin_last_block(ctrl) && threadIdx.x == 0)
+      run_ctrl_tail(CF, ctrl, nullptr);
+  }
 }
 
 // delta = -z + beta * delta
@@ -567,11 +703,13 @@ __global__ void k_axpby(const double* __restrict__ A,
 }
 
 // dots: ctrl[slot] += <A, B>, optionally ctrl[slot2] += <A, C>
+template <int CF = CF_NONE>
 __global__ void k_dots(const double* __restrict__ A,
                        const double* __restrict__ B,
                        const double* __restrict__ C,
                        double* __restrict__ ctrl,
-                       int slot, int slot2, long total, int guard) {
+                       int slot, int slot2, long total, int guard,
+                       double* __restrict__ cf_out = nullptr) {
   if (guarded_off(ctrl, guard)) return;
   const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   double d0 = 0.0, d1 = 0.0;
@@ -582,6 +720,10 @@ __global__ void k_dots(const double* __restrict__ A,
   }
   block_reduce_atomic(d0, ctrl + slot);
   if (C && slot2 >= 0) block_reduce_atomic(d1, ctrl + slot2);
+  if (CF != CF_NONE) {
+    if (fanin_last_block(ctrl) && threadIdx.x == 0)
+      run_ctrl_tail(CF, ctrl, cf_out);
+  }
 }
 
 // ---------------------------------------------------------------------
@@ -904,22 +1046,32 @@ static void launch_spmm(const int* rp, const int* ci, const double* vals,
                      rp, ci, vals, X, out, n, dh, r, ctrl, guard);
 }
 
-static void launch_proj_dots(const double* X, const double* V,
-                             const double* G, double* out,
-                             const double* dotWith, double* ctrl,
-                             int n, int d, int r, int dot_slot,
-                             int dot_slot2, int guard, hipStream_t s) {
+template <int CF>
+static void launch_proj_dots_cf(const double* X, const double* V,
+                                const double* G, double* out,
+                                const double* dotWith, double* ctrl,
+                                int n, int d, int r, int dot_slot,
+                                int dot_slot2, int guard, hipStream_t s) {
   const int grid = blocks_for(n, 256);
 #define CASE_PROJ(D, R) \
   if (d == D && r == R) { \
-    hipLaunchKernelGGL((k_proj_dots<0, D, R>), dim3(grid), dim3(256), 0, s, \
-                       X, V, G, out, dotWith, ctrl, n, dot_slot, \
+    hipLaunchKernelGGL((k_proj_dots<0, D, R, CF>), dim3(grid), dim3(256), \
+                       0, s, X, V, G, out, dotWith, ctrl, n, dot_slot, \
                        dot_slot2, guard); \
     return; \
   }
   DPO_FOREACH_DR(CASE_PROJ)
 #undef CASE_PROJ
   dpo_bad_shape(d, r);
+}
+
+static void launch_proj_dots(const double* X, const double* V,
+                             const double* G, double* out,
+                             const double* dotWith, double* ctrl,
+                             int n, int d, int r, int dot_slot,
+                             int dot_slot2, int guard, hipStream_t s) {
+  launch_proj_dots_cf<CF_NONE>(X, V, G, out, dotWith, ctrl, n, d, r,
+                               dot_slot, dot_slot2, guard, s);
 }
 
 static void launch_polar(const double* A, const double* B, const double* C,
@@ -1017,9 +1169,9 @@ void dpo_tcg_update(double* eta, double* rvec, const double* delta,
                     const double* Hd, double* eta_snap, double* delta_snap,
                     double* ctrl, long total, void* stream) {
   hipStream_t s = (hipStream_t)stream;
-  hipLaunchKernelGGL(k_tcg_update, dim3(blocks_for(total, 256)), dim3(256),
-                     0, s, eta, rvec, delta, Hd, eta_snap, delta_snap,
-                     ctrl, total);
+  hipLaunchKernelGGL((k_tcg_update<CF_NONE>), dim3(blocks_for(total, 256)),
+                     dim3(256), 0, s, eta, rvec, delta, Hd, eta_snap,
+                     delta_snap, ctrl, total);
 }
 
 void dpo_tcg_delta(double* delta, const double* z, double* ctrl, long total,
@@ -1048,8 +1200,9 @@ void dpo_dots(const double* A, const double* B, const double* C,
               double* ctrl, int slot, int slot2, long total, int guard,
               void* stream) {
   hipStream_t s = (hipStream_t)stream;
-  hipLaunchKernelGGL(k_dots, dim3(blocks_for(total, 256)), dim3(256),
-                     0, s, A, B, C, ctrl, slot, slot2, total, guard);
+  hipLaunchKernelGGL((k_dots<CF_NONE>), dim3(blocks_for(total, 256)),
+                     dim3(256), 0, s, A, B, C, ctrl, slot, slot2, total,
+                     guard, (double*)nullptr);
 }
 
 void dpo_ctrl_init(double* ctrl, double tol, double Delta0, double theta,
@@ -1321,20 +1474,19 @@ static void enqueue_solve_body(DpoCtx* c, double* X, const double* nbr,
   hipLaunchKernelGGL(k_tcg_delta, dim3(gvec), dim3(256), 0, s,
                      c->delta, c->z, c->ctrl, total);
 
-  // tCG loop
+  // tCG loop — the per-iteration control logic (alpha/boundary, rr
+  // convergence, beta) runs as fused fan-in tails of the reduction
+  // kernels (6 kernels per iteration instead of 9).
   for (int j = 0; j < c->max_inner; ++j) {
     ctx_spmm(c, c->delta, c->Hd, ST_RUN, s);
-    launch_proj_dots(X, c->Hd, nullptr, c->Hd, c->delta, c->ctrl, n, d,
-                     r, C_DOT0, -1, ST_RUN, s);
-    hipLaunchKernelGGL(k_ctrl_alpha, dim3(1), dim3(64), 0, s, c->ctrl);
-    hipLaunchKernelGGL(k_tcg_update, dim3(gvec), dim3(256), 0, s,
+    launch_proj_dots_cf<CF_ALPHA>(X, c->Hd, nullptr, c->Hd, c->delta,
+                                  c->ctrl, n, d, r, C_DOT0, -1, ST_RUN, s);
+    hipLaunchKernelGGL((k_tcg_update<CF_RR>), dim3(gvec), dim3(256), 0, s,
                        c->eta, c->rvec, c->delta, c->Hd, c->eta_snap,
                        c->delta_snap, c->ctrl, total);
-    hipLaunchKernelGGL(k_ctrl_rr, dim3(1), dim3(64), 0, s, c->ctrl);
     ctx_precond(c, c->rvec, c->z, s);
-    launch_proj_dots(X, c->z, nullptr, c->z, c->rvec, c->ctrl, n, d, r,
-                     C_DOT0, -1, ST_RUN, s);
-    hipLaunchKernelGGL(k_ctrl_beta, dim3(1), dim3(64), 0, s, c->ctrl);
+    launch_proj_dots_cf<CF_BETA>(X, c->z, nullptr, c->z, c->rvec, c->ctrl,
+                                 n, d, r, C_DOT0, -1, ST_RUN, s);
     hipLaunchKernelGGL(k_tcg_delta, dim3(gvec), dim3(256), 0, s,
                        c->delta, c->z, c->ctrl, total);
   }
