@@ -1474,19 +1474,26 @@ static void enqueue_solve_body(DpoCtx* c, double* X, const double* nbr,
   hipLaunchKernelGGL(k_tcg_delta, dim3(gvec), dim3(256), 0, s,
                      c->delta, c->z, c->ctrl, total);
 
-  // tCG loop — the per-iteration control logic (alpha/boundary, rr
-  // convergence, beta) runs as fused fan-in tails of the reduction
-  // kernels (6 kernels per iteration instead of 9).
+  // tCG loop. NOTE: a fused fan-in variant (control logic as split-K
+  // tails of the reduction kernels, CF_* templates above) measured +8%
+  // throughput but showed intermittent run-to-run trajectory scatter on
+  // full convergence runs; until that is root-caused the control
+  // decisions run as dedicated single-wave kernels (each kernel
+  // boundary is an agent-scope acquire, making the cross-workgroup
+  // hand-off trivially correct).
   for (int j = 0; j < c->max_inner; ++j) {
     ctx_spmm(c, c->delta, c->Hd, ST_RUN, s);
-    launch_proj_dots_cf<CF_ALPHA>(X, c->Hd, nullptr, c->Hd, c->delta,
-                                  c->ctrl, n, d, r, C_DOT0, -1, ST_RUN, s);
-    hipLaunchKernelGGL((k_tcg_update<CF_RR>), dim3(gvec), dim3(256), 0, s,
-                       c->eta, c->rvec, c->delta, c->Hd, c->eta_snap,
+    launch_proj_dots(X, c->Hd, nullptr, c->Hd, c->delta, c->ctrl, n, d,
+                     r, C_DOT0, -1, ST_RUN, s);
+    hipLaunchKernelGGL(k_ctrl_alpha, dim3(1), dim3(64), 0, s, c->ctrl);
+    hipLaunchKernelGGL((k_tcg_update<CF_NONE>), dim3(gvec), dim3(256), 0,
+                       s, c->eta, c->rvec, c->delta, c->Hd, c->eta_snap,
                        c->delta_snap, c->ctrl, total);
+    hipLaunchKernelGGL(k_ctrl_rr, dim3(1), dim3(64), 0, s, c->ctrl);
     ctx_precond(c, c->rvec, c->z, s);
-    launch_proj_dots_cf<CF_BETA>(X, c->z, nullptr, c->z, c->rvec, c->ctrl,
-                                 n, d, r, C_DOT0, -1, ST_RUN, s);
+    launch_proj_dots(X, c->z, nullptr, c->z, c->rvec, c->ctrl, n, d, r,
+                     C_DOT0, -1, ST_RUN, s);
+    hipLaunchKernelGGL(k_ctrl_beta, dim3(1), dim3(64), 0, s, c->ctrl);
     hipLaunchKernelGGL(k_tcg_delta, dim3(gvec), dim3(256), 0, s,
                        c->delta, c->z, c->ctrl, total);
   }
