@@ -713,6 +713,29 @@ class DistributedRBCDDriver:
         self._sync_anchor()
         return res
 
+    def gather_final_trajectory(self):
+        """Rounded global trajectory (d, (d+1) n) on rank 0 (reference
+        PartitionInitial.cpp:329-335 output). Returns None on other
+        ranks."""
+        import torch
+        self._sync_anchor()
+        dh = self.dh
+        buf = torch.zeros(self.n_global, self.d, dh, dtype=torch.float64)
+        for rb, a in self.local_agents.items():
+            T = a.get_trajectory_in_global_frame()
+            if T is None:
+                continue
+            Tb = T.reshape(self.d, a.n, dh).transpose(1, 0, 2)
+            for i in range(a.n):
+                buf[self.pose_to_index[(rb, i)]] = torch.from_numpy(
+                    np.ascontiguousarray(Tb[i]))
+        self.comm.all_reduce_sum_(buf.view(-1))
+        if self.comm.rank != 0:
+            return None
+        return np.ascontiguousarray(
+            buf.numpy().transpose(1, 0, 2).reshape(self.d,
+                                                   self.n_global * dh))
+
     def _sync_anchor(self):
         import torch
         blk = self.dh * self.r

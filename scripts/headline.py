@@ -66,13 +66,14 @@ def main():
         "ms_per_iter": res.elapsed_s / max(res.iterations, 1) * 1e3,
     }
     rank = int(os.environ.get("RANK", "0"))
-    if args.dump_trajectory and args.driver == "local" and rank == 0:
-        import numpy as np
+    if args.dump_trajectory:
         from dpo_amd.logger import PGOLogger
-        T = drv.final_trajectory()
-        lg = PGOLogger(os.path.dirname(args.dump_trajectory) or ".")
-        lg.log_trajectory(meas[0].d, n, T,
-                          os.path.basename(args.dump_trajectory))
+        T = (drv.final_trajectory() if args.driver == "local"
+             else drv.gather_final_trajectory())
+        if T is not None and rank == 0:
+            lg = PGOLogger(os.path.dirname(args.dump_trajectory) or ".")
+            lg.log_trajectory(meas[0].d, n, T,
+                              os.path.basename(args.dump_trajectory))
     if rank == 0:
         print(json.dumps(out))
 
