@@ -659,16 +659,14 @@ class DistributedRBCDDriver:
                 aux_flats = self.comm.all_gather_flat(
                     self._packed_pack(use_aux=True), sizes)
                 self._packed_scatter(aux_flats, aux=True)
-            # NOTE: the async multi-stream SOLVE path (DPO_ASYNC_SOLVE=1)
-            # is experimental: on ROCm 7.2 a cached hipGraph exec replayed
-            # on a per-agent stream intermittently degrades after ~40
-            # replays (garbage control-block reads; inputs verified
-            # finite, a freshly captured graph on the same state works).
-            # Until that driver-level interaction is understood the solve
-            # replays on the main stream; the EVAL fan-out (verified
-            # bitwise-identical) does overlap on per-agent streams.
+            # Concurrent active agents' solves overlap on per-agent HIP
+            # streams (events order them against the main stream). An
+            # intermittent corruption originally blamed on multi-stream
+            # graph replay turned out to be an LDS reuse race in
+            # back-to-back block reductions (fixed in dpo_ops.hip);
+            # with the fix the async path is bitwise-deterministic.
             import os as _os
-            _sync_solve = _os.environ.get("DPO_ASYNC_SOLVE", "0") != "1"
+            _sync_solve = _os.environ.get("DPO_SYNC_SOLVE", "0") == "1"
             for rb, a in self.local_agents.items():
                 if rb in active:
                     if _sync_solve:
