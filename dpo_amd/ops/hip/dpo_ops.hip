@@ -401,6 +401,127 @@ __global__ void k_proj_dots(const double* __restrict__ X,
 }
 
 // ---------------------------------------------------------------------
+// Fused Hessian-vector product: one THREAD per pose computes its whole
+// BSR row of Q @ V (the dh x r accumulator tile lives in registers; each
+// Q block is one 128-byte line), optionally adds G, applies the tangent
+// projection at X, writes the result and accumulates the requested dot
+// products — replacing a {SpMM kernel, projection kernel} pair with one
+// launch. MODE 0: out = P_X(QV + G), dots <out, dotW> / <QV+G, X>.
+// MODE 1 (no projection, no store): dots <QV, V> and <G, V> (the
+// trust-region candidate's f evaluation).
+// ---------------------------------------------------------------------
+template <int D, int R, int MODE>
+__global__ void k_hess_fused(const int* __restrict__ row_ptr,
+                             const int* __restrict__ col_idx,
+                             const double* __restrict__ vals,
+                             const double* __restrict__ V,
+                             const double* __restrict__ X,
+                             const double* __restrict__ G,
+                             double* __restrict__ out,
+                             const double* __restrict__ dotW,
+                             double* __restrict__ ctrl,
+                             int n, int dot_slot, int dot_slot2,
+                             int guard) {
+  if (guarded_off(ctrl, guard)) return;
+  constexpr int dh = D + 1;
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  double acc[dh][R];
+  double d0 = 0.0, d1 = 0.0;
+  if (i < n) {
+    #pragma unroll
+    for (int c = 0; c < dh; ++c)
+      #pragma unroll
+      for (int k = 0; k < R; ++k) acc[c][k] = 0.0;
+    const int s0 = row_ptr[i], e0 = row_ptr[i + 1];
+    for (int p = s0; p < e0; ++p) {
+      const double* B = vals + (size_t)p * dh * dh;
+      const double* Vj = V + (size_t)col_idx[p] * dh * R;
+      #pragma unroll
+      for (int c = 0; c < dh; ++c)
+        #pragma unroll
+        for (int cc = 0; cc < dh; ++cc) {
+          const double b = B[c * dh + cc];
+          #pragma unroll
+          for (int k = 0; k < R; ++k)
+            acc[c][k] = fma(b, Vj[cc * R + k], acc[c][k]);
+        }
+    }
+    if (G) {
+      const double* Gi = G + (size_t)i * dh * R;
+      #pragma unroll
+      for (int c = 0; c < dh; ++c)
+        #pragma unroll
+        for (int k = 0; k < R; ++k) acc[c][k] += Gi[c * R + k];
+    }
+    if (MODE == 1) {
+      // dots only: d0 = <(QV)_i + G_i, V_i>, d1 = <G_i, V_i>
+      const double* Vi = V + (size_t)i * dh * R;
+      const double* Gi = G ? G + (size_t)i * dh * R : nullptr;
+      #pragma unroll
+      for (int c = 0; c < dh; ++c)
+        #pragma unroll
+        for (int k = 0; k < R; ++k) {
+          const double g = Gi ? Gi[c * R + k] : 0.0;
+          d0 = fma(acc[c][k] - g, Vi[c * R + k], d0);
+          d1 = fma(g, Vi[c * R + k], d1);
+        }
+    } else {
+      const double* Xi = X + (size_t)i * dh * R;
+      if (dot_slot2 >= 0) {
+        #pragma unroll
+        for (int c = 0; c < dh; ++c)
+          #pragma unroll
+          for (int k = 0; k < R; ++k)
+            d1 = fma(acc[c][k], Xi[c * R + k], d1);
+      }
+      // tangent projection at X: S = sym(Yt A^T); A -= S Yt
+      double S[D][D];
+      #pragma unroll
+      for (int a = 0; a < D; ++a)
+        #pragma unroll
+        for (int b = 0; b < D; ++b) {
+          double sv = 0.0;
+          #pragma unroll
+          for (int k = 0; k < R; ++k)
+            sv = fma(Xi[a * R + k], acc[b][k], sv);
+          S[a][b] = sv;
+        }
+      #pragma unroll
+      for (int a = 0; a < D; ++a)
+        #pragma unroll
+        for (int b = a; b < D; ++b) {
+          const double sv = 0.5 * (S[a][b] + S[b][a]);
+          S[a][b] = sv;
+          S[b][a] = sv;
+        }
+      #pragma unroll
+      for (int a = 0; a < D; ++a)
+        #pragma unroll
+        for (int k = 0; k < R; ++k) {
+          double v = acc[a][k];
+          #pragma unroll
+          for (int b = 0; b < D; ++b) v = fma(-S[a][b], Xi[b * R + k], v);
+          acc[a][k] = v;
+        }
+      double* Oi = out + (size_t)i * dh * R;
+      const double* Wi = dotW ? dotW + (size_t)i * dh * R : nullptr;
+      #pragma unroll
+      for (int c = 0; c < dh; ++c)
+        #pragma unroll
+        for (int k = 0; k < R; ++k) {
+          Oi[c * R + k] = acc[c][k];
+          if (dot_slot >= 0) {
+            const double w = Wi ? Wi[c * R + k] : acc[c][k];
+            d0 = fma(acc[c][k], w, d0);
+          }
+        }
+    }
+  }
+  if (dot_slot >= 0) block_reduce_atomic(d0, ctrl + dot_slot);
+  if (dot_slot2 >= 0) block_reduce_atomic(d1, ctrl + dot_slot2);
+}
+
+// ---------------------------------------------------------------------
 // Batched polar projection onto (St(d, r) x R^r)^n of an affine
 // combination  M = ca*A + cb*B + cc*C  (B, C optional).
 // polar(Mt) for the wide d x r Stiefel block via the analytic
@@ -1090,6 +1211,27 @@ static void launch_polar(const double* A, const double* B, const double* C,
   dpo_bad_shape(d, r);
 }
 
+template <int MODE>
+static void launch_hess_fused(const int* rp, const int* ci,
+                              const double* vals, const double* V,
+                              const double* X, const double* G,
+                              double* out, const double* dotW,
+                              double* ctrl, int n, int d, int r,
+                              int dot_slot, int dot_slot2, int guard,
+                              hipStream_t s) {
+  const int grid = blocks_for(n, 256);
+#define CASE_HF(D, R) \
+  if (d == D && r == R) { \
+    hipLaunchKernelGGL((k_hess_fused<D, R, MODE>), dim3(grid), dim3(256), \
+                       0, s, rp, ci, vals, V, X, G, out, dotW, ctrl, n, \
+                       dot_slot, dot_slot2, guard); \
+    return; \
+  }
+  DPO_FOREACH_DR(CASE_HF)
+#undef CASE_HF
+  dpo_bad_shape(d, r);
+}
+
 static void launch_precond_dense(const float* Minv, const double* V,
                                  double* Z, int N, int r,
                                  const double* ctrl, int guard,
@@ -1482,9 +1624,9 @@ static void enqueue_solve_body(DpoCtx* c, double* X, const double* nbr,
   // boundary is an agent-scope acquire, making the cross-workgroup
   // hand-off trivially correct).
   for (int j = 0; j < c->max_inner; ++j) {
-    ctx_spmm(c, c->delta, c->Hd, ST_RUN, s);
-    launch_proj_dots(X, c->Hd, nullptr, c->Hd, c->delta, c->ctrl, n, d,
-                     r, C_DOT0, -1, ST_RUN, s);
+    launch_hess_fused<0>(c->q_rp, c->q_ci, c->q_vals, c->delta, X,
+                         nullptr, c->Hd, c->delta, c->ctrl, n, d, r,
+                         C_DOT0, -1, ST_RUN, s);
     hipLaunchKernelGGL(k_ctrl_alpha, dim3(1), dim3(64), 0, s, c->ctrl);
     hipLaunchKernelGGL((k_tcg_update<CF_NONE>), dim3(gvec), dim3(256), 0,
                        s, c->eta, c->rvec, c->delta, c->Hd, c->eta_snap,
@@ -1505,10 +1647,9 @@ static void enqueue_solve_body(DpoCtx* c, double* X, const double* nbr,
                      c->ctrl, total);
   launch_polar(X, c->step, nullptr, 1.0, 1.0, 0.0, c->Xprop, n, d, r,
                c->ctrl, ST_TCG_STOP, s);
-  ctx_spmm(c, c->Xprop, c->W, ST_TCG_STOP, s);
-  hipLaunchKernelGGL(k_dots, dim3(gvec), dim3(256), 0, s,
-                     c->Xprop, c->W, c->Gt, c->ctrl, C_DOT0, C_DOT2,
-                     total, ST_TCG_STOP);
+  launch_hess_fused<1>(c->q_rp, c->q_ci, c->q_vals, c->Xprop, nullptr,
+                       c->Gt, nullptr, nullptr, c->ctrl, n, d, r,
+                       C_DOT0, C_DOT2, ST_TCG_STOP, s);
   hipLaunchKernelGGL(k_ctrl_accept, dim3(1), dim3(64), 0, s, c->ctrl,
                      accept_rho);
   DPO_CHECK(hipMemcpyAsync(c->ctrl_host, c->ctrl,
@@ -1593,10 +1734,9 @@ static int solve_postsync(DpoCtx* c, double* X, double accept_rho,
                          c->ctrl, total);
       launch_polar(X, c->step, nullptr, 1.0, 1.0, 0.0, c->Xprop, n, d, r,
                    c->ctrl, ST_TCG_STOP, s);
-      ctx_spmm(c, c->Xprop, c->W, ST_TCG_STOP, s);
-      hipLaunchKernelGGL(k_dots, dim3(gvec), dim3(256), 0, s,
-                         c->Xprop, c->W, c->Gt, c->ctrl, C_DOT0, C_DOT2,
-                         total, ST_TCG_STOP);
+      launch_hess_fused<1>(c->q_rp, c->q_ci, c->q_vals, c->Xprop, nullptr,
+                           c->Gt, nullptr, nullptr, c->ctrl, n, d, r,
+                           C_DOT0, C_DOT2, ST_TCG_STOP, s);
       hipLaunchKernelGGL(k_ctrl_accept, dim3(1), dim3(64), 0, s, c->ctrl,
                          accept_rho);
       DPO_CHECK(hipMemcpyAsync(c->ctrl_host, c->ctrl,
@@ -1621,9 +1761,9 @@ static int solve_postsync(DpoCtx* c, double* X, double accept_rho,
   double gn_opt = gn_init;
   if (compute_final_gn && status == ST_ACCEPTED) {
     DPO_CHECK(hipMemsetAsync(c->ctrl + C_DOT1, 0, sizeof(double), s));
-    ctx_spmm(c, X, c->W, -1, s);
-    launch_proj_dots(X, c->W, c->Gt, c->grad, nullptr, c->ctrl, n, d, r,
-                     C_DOT1, -1, -1, s);
+    launch_hess_fused<0>(c->q_rp, c->q_ci, c->q_vals, X, X, c->Gt,
+                         c->grad, nullptr, c->ctrl, n, d, r, C_DOT1, -1,
+                         -1, s);
     DPO_CHECK(hipMemcpyAsync(c->ctrl_host + C_DOT1, c->ctrl + C_DOT1,
                              sizeof(double), hipMemcpyDeviceToHost, s));
     DPO_CHECK(hipStreamSynchronize(s));
