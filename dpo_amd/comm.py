@@ -13,7 +13,7 @@ from __future__ import annotations
 
 import datetime
 import os
-from typing import List, Optional
+from typing import List
 
 import torch
 import torch.distributed as dist

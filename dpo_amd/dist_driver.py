@@ -21,7 +21,6 @@ with world_size == 1 (all agents on one device).
 from __future__ import annotations
 
 import time
-from dataclasses import dataclass, field
 from typing import Dict, List, Optional, Sequence, Tuple
 
 import numpy as np
@@ -583,7 +582,6 @@ class DistributedRBCDDriver:
 
     def _run_packed(self, max_iters, gradnorm_tol, trace_file, time_limit_s):
         import torch
-        from .types import OptAlgorithm
         res = RBCDResult()
         self._packed_setup()
         pk = self._pk

@@ -11,7 +11,7 @@ BASELINE.md (e.g. city10000 naive 33448 -> ~260 at k=5).
 from __future__ import annotations
 
 import heapq
-from typing import Dict, List, Optional, Sequence, Tuple
+from typing import Dict, List, Sequence, Tuple
 
 import numpy as np
 

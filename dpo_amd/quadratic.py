@@ -23,7 +23,7 @@ rebuild values by scaling — a scatter-add kernel on GPU.
 from __future__ import annotations
 
 from dataclasses import dataclass
-from typing import Dict, List, Optional, Sequence, Tuple
+from typing import Optional, Sequence, Tuple
 
 import numpy as np
 import torch
