@@ -417,6 +417,50 @@ class PGOAgent:
         T_opt[:self.d, self.d] = t_opt
         return T_opt
 
+    def _compute_robust_neighbor_transform(
+            self, neighbor_id: int,
+            pose_dict: Dict[PoseID, np.ndarray]) -> np.ndarray:
+        """Single-stage GNC pose averaging over candidate alignments
+        (reference PGOAgent.cpp:333-367; the two-stage variant below is
+        what initializeInGlobalFrame uses by default)."""
+        from .averaging import robust_single_pose_averaging
+        from .robust import RobustCost
+        R_vec, t_vec = [], []
+        for nid, var in pose_dict.items():
+            if nid in self.neighbor_shared_pose_ids:
+                T = self._compute_neighbor_transform(nid, np.asarray(var))
+                R_vec.append(T[:self.d, :self.d])
+                t_vec.append(T[:self.d, self.d])
+        if not R_vec:
+            raise RuntimeError("no shared edges with neighbor")
+        kappa = 1.82 * np.ones(len(R_vec))   # rot stddev ~30 deg
+        tau = 0.01 * np.ones(len(R_vec))     # trans stddev ~10 m
+        cbar = RobustCost.error_threshold_at_quantile(0.9, 3)
+        R_opt, t_opt, inliers = robust_single_pose_averaging(
+            R_vec, t_vec, kappa, tau, cbar)
+        if len(inliers) == 0:
+            raise RuntimeError("empty inlier set in robust initialization")
+        T_opt = np.eye(self.dh)
+        T_opt[:self.d, :self.d] = R_opt
+        T_opt[:self.d, self.d] = t_opt
+        return T_opt
+
+    def get_aux_shared_pose(self, index: int) -> Optional[np.ndarray]:
+        """Nesterov auxiliary pose block (reference PGOAgent.cpp:85-93)."""
+        assert self.params.acceleration
+        if self.state != PGOAgentState.INITIALIZED or index >= self.n:
+            return None
+        with self._lock:
+            return self._block(self.Y, index)
+
+    @staticmethod
+    def is_duplicate_measurement(m: RelativeSEMeasurement,
+                                 measurements) -> bool:
+        """Reference PGOAgent.cpp:1291-1299."""
+        return any(m.r1 == m2.r1 and m.r2 == m2.r2
+                   and m.p1 == m2.p1 and m.p2 == m2.p2
+                   for m2 in measurements)
+
     def initialize_in_global_frame(self, neighbor_id: int,
                                    pose_dict: Dict[PoseID, np.ndarray]) -> None:
         assert self.YLift is not None
