@@ -576,7 +576,8 @@ class PGOAgent:
                                                 self.device, max_inner=10)
             X_work = X_start.clone() if acceleration else self.X
             stats = self._dev_solver.solve(self.problem, X_work,
-                                           tol=1e-2, Delta0=100.0)
+                                           tol=self.params.inner_tol,
+                                           Delta0=100.0)
             self.X = X_work
             from .types import OptResult
             res = OptResult(success=True,
@@ -585,7 +586,7 @@ class PGOAgent:
                             grad_norm_opt=stats.get("grad_norm_opt", 0.0))
             self.last_opt_result = res
             return True
-        tr = TRParams(tolerance=1e-2, initial_radius=100.0,
+        tr = TRParams(tolerance=self.params.inner_tol, initial_radius=100.0,
                       max_iterations=1, max_inner_iterations=10)
         opt = QuadraticOptimizer(self.problem, self.params.algorithm, tr,
                                  verbose=self.params.verbose)
@@ -966,7 +967,8 @@ class PGOAgent:
         if accel:
             self.X.copy_(self.Y)
         nbr = self._nbr_buffer_aux if accel else self._nbr_buffer
-        self._dev_solver.round_solve(self.X, nbr, tol=1e-2, Delta0=100.0)
+        self._dev_solver.round_solve(self.X, nbr, tol=self.params.inner_tol,
+                                     Delta0=100.0)
 
     def _packed_eval(self, out=None):
         """Device 3-vector [f, 0.5<X,G>, ||rgrad||^2] with fresh G.
@@ -981,7 +983,8 @@ class PGOAgent:
         if accel:
             self.X.copy_(self.Y)
         nbr = self._nbr_buffer_aux if accel else self._nbr_buffer
-        self._dev_solver.round_solve_async(self.X, nbr)
+        self._dev_solver.round_solve_async(self.X, nbr,
+                                           tol=self.params.inner_tol)
 
     def _packed_solve_finish(self) -> None:
         self._dev_solver.round_solve_finish(self.X)
@@ -1111,7 +1114,8 @@ class PGOAgent:
             # periodic restart (PGOAgent.cpp:1040-1052)
             self.X.copy_(self._XPrev_packed)
             self._dev_solver.round_solve(self.X, self._nbr_buffer,
-                                         tol=1e-2, Delta0=100.0)
+                                         tol=self.params.inner_tol,
+                                         Delta0=100.0)
             self.V.copy_(self.X)
             self.Y.copy_(self.X)
             self.gamma = 0.0

@@ -61,7 +61,8 @@ class DistributedRBCDDriver:
                  robust: RobustCostType = RobustCostType.L2,
                  device: str = "cpu",
                  verbose: bool = False,
-                 selection: str = "greedy"):
+                 selection: str = "greedy",
+                 inner_tol: float = 1e-2):
         self.comm = comm
         self.num_robots = num_robots
         self.verbose = verbose
@@ -151,7 +152,8 @@ class DistributedRBCDDriver:
             p = PGOAgentParams(d=d, r=r, num_robots=num_robots,
                                acceleration=acceleration,
                                robust_cost_type=robust,
-                               verbose=verbose, device=device)
+                               verbose=verbose, device=device,
+                               inner_tol=inner_tol)
             a = PGOAgent(rb, p)
             a.set_lifting_matrix(YL)
             T_init = None

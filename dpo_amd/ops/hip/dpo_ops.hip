@@ -26,7 +26,15 @@
 #include <math.h>
 #include <stdint.h>
 #include <stdio.h>
+#include <stdlib.h>
 #include <string.h>
+
+// Debug/bisection knobs: DPO_NO_SOLVE_GRAPH=1 / DPO_NO_EVAL_GRAPH=1 run
+// the solve / eval sequences eagerly instead of via cached hipGraphs.
+static bool dpo_env_flag(const char* name) {
+  const char* v = getenv(name);
+  return v && v[0] == '1';
+}
 
 #define DPO_CHECK(x)                                                     \
   do {                                                                   \
@@ -1454,7 +1462,8 @@ struct DpoCtx {
   long total;
   double *W, *grad, *eta, *delta, *rvec, *z, *Hd, *step, *Xprop;
   double *eta_snap, *delta_snap, *ctrl;
-  double *ctrl_host;  // pinned
+  double *ctrl_host;      // pinned
+  double *ctrl_host_dev;  // device-side view of ctrl_host (mapped)
   // borrowed problem pointers (owned by torch tensors on the Python side)
   const int *q_rp = nullptr, *q_ci = nullptr;
   const double *q_vals = nullptr;
@@ -1492,15 +1501,84 @@ struct DpoCtx {
   // state of an in-flight async solve
   double pend_tol = 0, pend_Delta0 = 0, pend_rho = 0;
   double* pend_X = nullptr;
+  // data-flow fence slots (see k_fence_* above)
+  unsigned int* fences = nullptr;
   void invalidate_graphs() {
     if (solve_graph) { hipGraphExecDestroy(solve_graph); solve_graph = nullptr; }
     if (eval_graph) { hipGraphExecDestroy(eval_graph); eval_graph = nullptr; }
   }
 };
 
+// --- data-flow fences around hipGraph launches ----------------------
+// ROCm 7.2 on gfx950: a hipGraphLaunch intermittently fails to order
+// against work enqueued on other streams via hipStreamWaitEvent (and,
+// on the legacy stream, against prior eager work), executing the graph
+// early/concurrently. Observed as rare stale/garbage reads in the
+// round pipeline (bisected: eager-everything is stable, every failing
+// configuration launches a graph whose cross-stream ordering is
+// load-bearing). Instead of depending on launch ordering at all, each
+// graph carries explicit data-flow fences:
+//   producer stream:  k_fence_signal(IN)   (eager kernel — reliable)
+//   graph first node: k_fence_wait(IN)     (spins until signalled)
+//   graph last node:  k_fence_signal(OUT)
+//   consumer stream:  k_fence_wait(OUT)    (eager kernel after launch)
+// so even a misordered launch blocks on its first node until its
+// inputs are ready, and downstream eager work blocks until the graph
+// really finished. Cost: ~2 one-wave kernels per launch (<2 us).
+enum { F_SOLVE_IN = 0, F_SOLVE_OUT = 1, F_EVAL_IN = 2, F_EVAL_OUT = 3,
+       F_NUM = 4 };
+
+__global__ void k_fence_signal(unsigned int* f) {
+  if (threadIdx.x == 0)
+    __hip_atomic_store(f, 1u, __ATOMIC_RELEASE, __HIP_MEMORY_SCOPE_AGENT);
+}
+
+__global__ void k_fence_wait(unsigned int* f) {
+  if (threadIdx.x == 0) {
+    long spins = 0;
+    while (__hip_atomic_load(f, __ATOMIC_ACQUIRE,
+                             __HIP_MEMORY_SCOPE_AGENT) == 0u) {
+      __builtin_amdgcn_s_sleep(64);
+      // ~50 s at ~1.7us/iter: a lost signal is a programming error —
+      // fail loudly (aborted kernel surfaces on the next sync) rather
+      // than silently proceeding on unordered data.
+      if (++spins > 30000000L) __builtin_trap();
+    }
+    __hip_atomic_store(f, 0u, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+  }
+}
+
+static inline void fence_signal(DpoCtx* c, int which, hipStream_t s) {
+  hipLaunchKernelGGL(k_fence_signal, dim3(1), dim3(64), 0, s,
+                     c->fences + which);
+}
+static inline void fence_wait(DpoCtx* c, int which, hipStream_t s) {
+  hipLaunchKernelGGL(k_fence_wait, dim3(1), dim3(64), 0, s,
+                     c->fences + which);
+}
+
+// Graph-safe zero / device->pinned-host copy. hipMemsetAsync and
+// hipMemcpyAsync become SDMA-engine nodes inside a captured hipGraph;
+// compute<->SDMA dependencies in replayed graphs are part of the same
+// ROCm 7.2 ordering fragility the fences work around, so the captured
+// bodies use plain kernels only.
+__global__ void k_dzero(double* p, long n) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) p[i] = 0.0;
+}
+__global__ void k_ctrl_to_host(const double* __restrict__ src,
+                               double* __restrict__ dst, int n) {
+  int i = threadIdx.x;
+  if (i < n) dst[i] = src[i];
+}
+static inline void dzero(double* p, long n, hipStream_t s) {
+  hipLaunchKernelGGL(k_dzero, dim3((unsigned)((n + 255) / 256)), dim3(256),
+                     0, s, p, n);
+}
+
 static void ctx_assemble_g(DpoCtx* c, const double* nbr, hipStream_t s) {
   if (c->g_ne == 0) { c->Gt = nullptr; return; }
-  DPO_CHECK(hipMemsetAsync(c->G_buf, 0, c->total * sizeof(double), s));
+  dzero(c->G_buf, c->total, s);
   hipLaunchKernelGGL(k_g_assemble,
                      dim3(((long)c->g_ne * c->dh * c->r + 255) / 256),
                      dim3(256), 0, s, c->G_buf, c->g_E0, c->g_local_pose,
@@ -1545,6 +1623,8 @@ void* dpo_ctx_create(int n, int d, int r, int max_inner) {
   DPO_CHECK(hipMalloc(&c->ctrl, CTRL_SIZE * sizeof(double)));
   DPO_CHECK(hipMalloc(&c->G_buf, vb));
   DPO_CHECK(hipHostMalloc(&c->ctrl_host, CTRL_SIZE * sizeof(double)));
+  DPO_CHECK(hipHostGetDevicePointer((void**)&c->ctrl_host_dev,
+                                    c->ctrl_host, 0));
   DPO_CHECK(hipStreamCreateWithFlags(&c->cap_stream,
                                      hipStreamNonBlocking));
   DPO_CHECK(hipStreamCreateWithFlags(&c->exec_stream,
@@ -1553,6 +1633,8 @@ void* dpo_ctx_create(int n, int d, int r, int max_inner) {
                                     hipEventDisableTiming));
   DPO_CHECK(hipEventCreateWithFlags(&c->done_event,
                                     hipEventDisableTiming));
+  DPO_CHECK(hipMalloc(&c->fences, F_NUM * sizeof(unsigned int)));
+  DPO_CHECK(hipMemset(c->fences, 0, F_NUM * sizeof(unsigned int)));
   return c;
 }
 
@@ -1567,6 +1649,7 @@ void dpo_ctx_destroy(void* h) {
   if (c->start_event) hipEventDestroy(c->start_event);
   if (c->done_event) hipEventDestroy(c->done_event);
   hipFree(c->ctrl); hipFree(c->G_buf); hipHostFree(c->ctrl_host);
+  hipFree(c->fences);
   delete c;
 }
 
@@ -1590,10 +1673,11 @@ static void enqueue_solve_body(DpoCtx* c, double* X, const double* nbr,
   const int n = c->n, d = c->d, r = c->r;
   const long total = c->total;
   const int gvec = (int)((total + 255) / 256);
+  fence_wait(c, F_SOLVE_IN, s);  // first node: block until inputs ready
   if (nbr) ctx_assemble_g(c, nbr, s);
-  DPO_CHECK(hipMemsetAsync(c->ctrl, 0, CTRL_SIZE * sizeof(double), s));
-  DPO_CHECK(hipMemsetAsync(c->eta, 0, total * sizeof(double), s));
-  DPO_CHECK(hipMemsetAsync(c->delta, 0, total * sizeof(double), s));
+  dzero(c->ctrl, CTRL_SIZE, s);
+  dzero(c->eta, total, s);
+  dzero(c->delta, total, s);
 
   // gradient phase
   ctx_spmm(c, X, c->W, -1, s);
@@ -1652,9 +1736,9 @@ static void enqueue_solve_body(DpoCtx* c, double* X, const double* nbr,
                        C_DOT0, C_DOT2, ST_TCG_STOP, s);
   hipLaunchKernelGGL(k_ctrl_accept, dim3(1), dim3(64), 0, s, c->ctrl,
                      accept_rho);
-  DPO_CHECK(hipMemcpyAsync(c->ctrl_host, c->ctrl,
-                           CTRL_SIZE * sizeof(double),
-                           hipMemcpyDeviceToHost, s));
+  hipLaunchKernelGGL(k_ctrl_to_host, dim3(1), dim3(64), 0, s, c->ctrl,
+                     c->ctrl_host_dev, CTRL_SIZE);
+  fence_signal(c, F_SOLVE_OUT, s);  // last node: mark sequence complete
 }
 
 // Full RBCD local solve in place on X. stats_out (host, >= 8 doubles):
@@ -1668,6 +1752,13 @@ static void enqueue_solve_body(DpoCtx* c, double* X, const double* nbr,
 static bool solve_presync(DpoCtx* c, double* X, const double* nbr,
                           double tol, double Delta0, double accept_rho,
                           hipStream_t s) {
+  static const bool no_graph = dpo_env_flag("DPO_NO_SOLVE_GRAPH");
+  if (no_graph) {
+    enqueue_solve_body(c, X, nbr, tol, Delta0, accept_rho, s);
+    fence_wait(c, F_SOLVE_OUT, s);
+    DPO_CHECK(hipStreamSynchronize(s));
+    return false;
+  }
   const void* key[4] = {X, nbr, (const void*)(intptr_t)(tol * 1e9),
                         (const void*)(intptr_t)Delta0};
   bool key_match = c->solve_graph && memcmp(key, c->solve_key,
@@ -1703,6 +1794,7 @@ static bool solve_presync(DpoCtx* c, double* X, const double* nbr,
       // capture unavailable: run eagerly on the caller's stream
       c->solve_graph = nullptr;
       enqueue_solve_body(c, X, nbr, tol, Delta0, accept_rho, s);
+      fence_wait(c, F_SOLVE_OUT, s);
       DPO_CHECK(hipStreamSynchronize(s));
       return false;
     }
@@ -1710,6 +1802,9 @@ static bool solve_presync(DpoCtx* c, double* X, const double* nbr,
   }
   c->solve_replays++;
   DPO_CHECK(hipGraphLaunch(c->solve_graph, s));
+  // eager wait pins downstream stream work (and host syncs) to the
+  // graph's actual completion even if the launch misordered
+  fence_wait(c, F_SOLVE_OUT, s);
   return true;
 }
 
@@ -1786,6 +1881,7 @@ static int rbcd_solve_impl(DpoCtx* c, double* X, const double* nbr,
                            double tol, double Delta0, int max_shrink,
                            double accept_rho, int compute_final_gn,
                            double* stats_out, hipStream_t s) {
+  fence_signal(c, F_SOLVE_IN, s);
   bool async = solve_presync(c, X, nbr, tol, Delta0, accept_rho, s);
   if (async) DPO_CHECK(hipStreamSynchronize(s));
   return solve_postsync(c, X, accept_rho, max_shrink, compute_final_gn,
@@ -1802,6 +1898,8 @@ int dpo_rbcd_solve(void* h, double* X, double tol, double Delta0,
 
 void dpo_round_eval(void* h, const double* X, const double* nbr,
                     double* out_dev, void* stream);  // fwd decl
+static void round_eval_impl(DpoCtx* c, const double* X, const double* nbr,
+                            double* out_dev, hipStream_t s);  // fwd decl
 
 // --- async (multi-stream) round entry points ------------------------
 // Launch the solve's pre-sync sequence on the ctx's private execution
@@ -1812,6 +1910,10 @@ void dpo_round_solve_async(void* h, double* X, const double* nbr,
                            void* join_stream) {
   DpoCtx* c = (DpoCtx*)h;
   hipStream_t js = (hipStream_t)join_stream;
+  // the IN fence is signalled on the PRODUCER (torch) stream: it
+  // carries the completion of the boundary-pose scatter into the solve
+  // sequence as a data dependency (see fence comment above)
+  fence_signal(c, F_SOLVE_IN, js);
   DPO_CHECK(hipEventRecord(c->start_event, js));
   DPO_CHECK(hipStreamWaitEvent(c->exec_stream, c->start_event, 0));
   solve_presync(c, X, nbr, tol, Delta0, accept_rho, c->exec_stream);
@@ -1836,9 +1938,11 @@ void dpo_round_eval_async(void* h, const double* X, const double* nbr,
                           double* out_dev, void* join_stream) {
   DpoCtx* c = (DpoCtx*)h;
   hipStream_t js = (hipStream_t)join_stream;
+  // IN fence on the producer stream: carries the boundary-pose scatter
+  fence_signal(c, F_EVAL_IN, js);
   DPO_CHECK(hipEventRecord(c->start_event, js));
   DPO_CHECK(hipStreamWaitEvent(c->exec_stream, c->start_event, 0));
-  dpo_round_eval(h, X, nbr, out_dev, (void*)c->exec_stream);
+  round_eval_impl(c, X, nbr, out_dev, c->exec_stream);
   DPO_CHECK(hipEventRecord(c->done_event, c->exec_stream));
 }
 
@@ -1856,7 +1960,7 @@ void dpo_eval_terms(void* h, const double* X, double* out_dev,
   const int n = c->n, d = c->d, r = c->r;
   const long total = c->total;
   const int gvec = (int)((total + 255) / 256);
-  DPO_CHECK(hipMemsetAsync(c->ctrl + C_DOT0, 0, 4 * sizeof(double), s));
+  dzero(c->ctrl + C_DOT0, 4, s);
   ctx_spmm(c, X, c->W, -1, s);
   launch_proj_dots(X, c->W, c->Gt, c->grad, nullptr, c->ctrl, n,
                    d, r, C_DOT1, C_DOT0, -1, s);
@@ -1890,6 +1994,7 @@ int dpo_round_solve(void* h, double* X, const double* nbr, double tol,
 
 // Non-caching eval enqueue (for embedding in an externally captured
 // graph, e.g. one driver-level graph covering every agent's eval).
+// Deliberately unfenced: the embedding graph owns the ordering.
 void dpo_round_eval_raw(void* h, const double* X, const double* nbr,
                         double* out_dev, void* stream) {
   DpoCtx* c = (DpoCtx*)h;
@@ -1897,10 +2002,25 @@ void dpo_round_eval_raw(void* h, const double* X, const double* nbr,
   dpo_eval_terms(h, X, out_dev, stream);
 }
 
-void dpo_round_eval(void* h, const double* X, const double* nbr,
-                    double* out_dev, void* stream) {
-  DpoCtx* c = (DpoCtx*)h;
-  hipStream_t s = (hipStream_t)stream;
+// Fenced eval body: wait on the IN fence, assemble G + evaluate, signal
+// the OUT fence. Used for both capture and eager enqueue.
+static void enqueue_eval_body(DpoCtx* c, const double* X, const double* nbr,
+                              double* out_dev, hipStream_t s) {
+  fence_wait(c, F_EVAL_IN, s);
+  ctx_assemble_g(c, nbr, s);
+  dpo_eval_terms((void*)c, X, out_dev, (void*)s);
+  fence_signal(c, F_EVAL_OUT, s);
+}
+
+// IN fence must already be signalled on the producer stream.
+static void round_eval_impl(DpoCtx* c, const double* X, const double* nbr,
+                            double* out_dev, hipStream_t s) {
+  static const bool no_graph = dpo_env_flag("DPO_NO_EVAL_GRAPH");
+  if (no_graph) {
+    enqueue_eval_body(c, X, nbr, out_dev, s);
+    fence_wait(c, F_EVAL_OUT, s);
+    return;
+  }
   const void* key[4] = {X, nbr, out_dev, nullptr};
   bool match = c->eval_graph && memcmp(key, c->eval_key, sizeof(key)) == 0
                && c->eval_replays < DpoCtx::kMaxReplays;
@@ -1915,8 +2035,7 @@ void dpo_round_eval(void* h, const double* X, const double* nbr,
     hipError_t rc = hipStreamBeginCapture(
         cs_, hipStreamCaptureModeThreadLocal);
     if (rc == hipSuccess) {
-      ctx_assemble_g(c, nbr, cs_);
-      dpo_eval_terms(h, X, out_dev, (void*)cs_);
+      enqueue_eval_body(c, X, nbr, out_dev, cs_);
       rc = hipStreamEndCapture(cs_, &graph);
     }
     hipStreamCaptureStatus capst = hipStreamCaptureStatusNone;
@@ -1933,14 +2052,23 @@ void dpo_round_eval(void* h, const double* X, const double* nbr,
     hipGetLastError();  // clear sticky capture-related error state
     if (rc != hipSuccess || !c->eval_graph) {
       c->eval_graph = nullptr;
-      ctx_assemble_g(c, nbr, s);
-      dpo_eval_terms(h, X, out_dev, stream);
+      enqueue_eval_body(c, X, nbr, out_dev, s);
+      fence_wait(c, F_EVAL_OUT, s);
       return;
     }
     memcpy(c->eval_key, key, sizeof(key));
   }
   c->eval_replays++;
   DPO_CHECK(hipGraphLaunch(c->eval_graph, s));
+  fence_wait(c, F_EVAL_OUT, s);
+}
+
+void dpo_round_eval(void* h, const double* X, const double* nbr,
+                    double* out_dev, void* stream) {
+  DpoCtx* c = (DpoCtx*)h;
+  hipStream_t s = (hipStream_t)stream;
+  fence_signal(c, F_EVAL_IN, s);
+  round_eval_impl(c, X, nbr, out_dev, s);
 }
 
 }  // extern "C"

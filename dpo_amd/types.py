@@ -136,6 +136,10 @@ class PGOAgentParams:
     robust_opt_min_convergence_ratio: float = 0.8
     max_num_iters: int = 500
     rel_change_tol: float = 5e-3
+    # Local trust-region solver gradient-norm tolerance (the reference
+    # hardwires 1e-2 in PGOAgent::optimize's RBCD knobs). 0.0 forces a
+    # full solve every round (benchmark mode: no converged no-op rounds).
+    inner_tol: float = 1e-2
     verbose: bool = False
     log_data: bool = False
     log_directory: str = ""
