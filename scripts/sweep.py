@@ -17,6 +17,8 @@ def main():
     ap.add_argument("--partition", default="contiguous")
     ap.add_argument("--robots", type=int, default=5)
     ap.add_argument("--max-iters", type=int, default=1000)
+    ap.add_argument("--datasets", default="",
+                    help="comma-separated subset (default: all)")
     args = ap.parse_args()
     from dpo_amd.comm import Comm
     from dpo_amd.dist_driver import DistributedRBCDDriver
@@ -24,6 +26,9 @@ def main():
     here = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
     names = sorted(os.path.splitext(os.path.basename(p))[0]
                    for p in glob.glob(os.path.join(here, "data", "*.npz")))
+    if args.datasets:
+        keep = {s.strip() for s in args.datasets.split(",")}
+        names = [n for n in names if n in keep]
     for name in names:
         try:
             meas, n = load_dataset(name)
