@@ -546,13 +546,19 @@ class DistributedRBCDDriver:
             # iteration-count scatter — parity with the reference's
             # deterministic trajectories wins. Opt in with
             # DPO_DRIVER_EVAL_GRAPH=1.
-            evalmat.zero_()
             _sync_eval_default = "0" if self.selection == "colored" else "1"
             if _os.environ.get("DPO_SYNC_EVAL", _sync_eval_default) == "1":
+                evalmat.zero_()
                 for rb, a in self.local_agents.items():
                     a._packed_eval(out=None)
                     evalmat[rb] = a._dev_solver._eval_out
+            elif getattr(self, "_group", None) is not None:
+                # native fan-out: one C call + one gather kernel
+                if not self._group_all:
+                    evalmat.zero_()
+                self._group.eval_all(evalmat)
             else:
+                evalmat.zero_()
                 for rb, a in self.local_agents.items():
                     a._packed_eval_async()
                 for rb, a in self.local_agents.items():
@@ -620,8 +626,76 @@ class DistributedRBCDDriver:
             self._round_counter = 0
         inner = next(iter(self.local_agents.values())).params \
             .robust_opt_inner_iters if self.local_agents else 30
+        # Pipelined evaluation readback: when no per-round host decision
+        # is needed (no greedy selection, no gradient-norm stop test),
+        # the per-round eval scalars land in a device-side ring and are
+        # read back every `chunk` rounds — the eval kernels of round t
+        # then overlap the solves of round t+1 instead of draining the
+        # GPU at a .cpu() sync every round.
+        import os as _os_
+        _sync_solve = _os_.environ.get("DPO_SYNC_SOLVE", "0") == "1"
+        _timing = _os_.environ.get("DPO_TIME_PHASES", "0") == "1"
+        ph = [0.0] * 5  # solve-enqueue, solve-finish, pack+scatter,
+        #                 eval-enqueue, ring-flush
+        # native multi-agent fan-out (one C call per phase, fixed
+        # pointers keep the per-agent hipGraph caches hot)
+        if (str(dev).startswith("cuda") and self.local_agents
+                and getattr(self, "_group", None) is None
+                and _os_.environ.get("DPO_NO_GROUP", "0") != "1"):
+            from .ops.hip_backend import DeviceGroup
+            rbs = sorted(self.local_agents)
+            self._group = DeviceGroup(
+                [self.local_agents[rb]._dev_solver for rb in rbs],
+                [self.local_agents[rb].X for rb in rbs],
+                [self.local_agents[rb]._nbr_buffer for rb in rbs],
+                rbs)
+            self._group_lidx = {rb: i for i, rb in enumerate(rbs)}
+            self._group_all = len(rbs) == self.num_robots
+        group = getattr(self, "_group", None)
+        use_group_solve = (group is not None and not accel
+                           and not _sync_solve)
+        inner_tol = next(iter(self.local_agents.values())) \
+            .params.inner_tol if self.local_agents else 1e-2
+        if self.selection == "colored":
+            color_active = [
+                [rb for rb in range(self.num_robots)
+                 if self._colors[rb] == c]
+                for c in range(self._num_colors)]
+            if group is not None:
+                color_ids = [
+                    group.ids([self._group_lidx[rb] for rb in ca
+                               if rb in self.local_agents])
+                    for ca in color_active]
+        chunk = 16 if (gradnorm_tol <= 0.0
+                       and self.selection != "greedy"
+                       and _os_.environ.get("DPO_DRIVER_EVAL_GRAPH",
+                                            "0") != "1") else 1
+        evalring = (torch.zeros(chunk, self.num_robots, 3,
+                                dtype=torch.float64, device=dev)
+                    if chunk > 1 else None)
+        pending = 0   # rounds evaluated but not yet read back
+
+        def flush_ring():
+            nonlocal pending
+            if pending == 0:
+                return
+            slab = evalring[:pending]
+            self.comm.all_reduce_sum_(slab)
+            evh = slab.cpu().numpy()
+            for k in range(pending):
+                c_ = float((evh[k, :, 0] - evh[k, :, 1]).sum())
+                g_ = float(np.sqrt(evh[k, :, 2].sum()))
+                res.trace.append((2.0 * c_, g_))
+                if fout:
+                    fout.write(f"{2.0 * c_:.10g},{g_:.10g}\n")
+            slab.zero_()
+            pending = 0
+
         for it in range(max_iters):
             self._round_counter += 1
+            if chunk > 1 and robust_mode \
+                    and self._round_counter % inner == 0:
+                flush_ring()
             if robust_mode and self._round_counter % inner == 0:
                 # GNC re-weighting round (reference iterate():
                 # shouldUpdateLoopClosureWeights -> update -> mu step)
@@ -644,9 +718,7 @@ class DistributedRBCDDriver:
                     a._packed_rebuild_q()
                     a.robust_cost.update()
             if self.selection == "colored":
-                color = it % self._num_colors
-                active = [rb for rb in range(self.num_robots)
-                          if self._colors[rb] == color]
+                active = color_active[it % self._num_colors]
             else:
                 active = [selected]
             if accel:
@@ -660,32 +732,68 @@ class DistributedRBCDDriver:
                     self._packed_pack(use_aux=True), sizes)
                 self._packed_scatter(aux_flats, aux=True)
             # Concurrent active agents' solves overlap on per-agent HIP
-            # streams (events order them against the main stream). An
-            # intermittent corruption originally blamed on multi-stream
-            # graph replay turned out to be an LDS reuse race in
-            # back-to-back block reductions (fixed in dpo_ops.hip);
-            # with the fix the async path is bitwise-deterministic.
-            import os as _os
-            _sync_solve = _os.environ.get("DPO_SYNC_SOLVE", "0") == "1"
-            for rb, a in self.local_agents.items():
-                if rb in active:
-                    if _sync_solve:
-                        a._packed_solve(accel)
-                    else:
-                        a._packed_solve_async(accel)
-                elif accel:
-                    a.X.copy_(a.Y)
-            if not _sync_solve:
-                for rb in active:
-                    if rb in self.local_agents:
-                        self.local_agents[rb]._packed_solve_finish()
+            # streams (data-flow fences in dpo_ops.hip order the cached
+            # solve/eval graphs against this stream's pack/scatter; the
+            # async path is bitwise-deterministic).
+            if _timing:
+                tA = time.perf_counter()
+            if use_group_solve:
+                if self.selection == "colored":
+                    gids = color_ids[it % self._num_colors]
+                elif selected in self.local_agents:
+                    gids = group.ids([self._group_lidx[selected]])
+                else:
+                    gids = group.ids([])
+                if len(gids):
+                    group.solve_start(gids, tol=inner_tol)
+                if _timing:
+                    tB = time.perf_counter()
+                if len(gids):
+                    group.solve_finish(gids)
+            else:
+                for rb, a in self.local_agents.items():
+                    if rb in active:
+                        if _sync_solve:
+                            a._packed_solve(accel)
+                        else:
+                            a._packed_solve_async(accel)
+                    elif accel:
+                        a.X.copy_(a.Y)
+                if _timing:
+                    tB = time.perf_counter()
+                if not _sync_solve:
+                    for rb in active:
+                        if rb in self.local_agents:
+                            self.local_agents[rb]._packed_solve_finish()
+            if _timing:
+                tC = time.perf_counter()
             if accel:
                 for a in self.local_agents.values():
                     a._packed_nesterov_post(it)
             flats = self.comm.all_gather_flat(self._packed_pack(), sizes)
             self._packed_scatter(flats)
+            if _timing:
+                tD = time.perf_counter()
+                ph[0] += tB - tA
+                ph[1] += tC - tB
+                ph[2] += tD - tC
             # evaluation (fresh neighbor data); agents fan out on their
             # own streams and join back before the packed reduce
+            if chunk > 1:
+                self._packed_eval_phase(evalring[pending])
+                pending += 1
+                res.iterations = it + 1
+                if _timing:
+                    ph[3] += time.perf_counter() - tD
+                if pending == chunk or it + 1 == max_iters:
+                    tE = time.perf_counter()
+                    flush_ring()
+                    if _timing:
+                        ph[4] += time.perf_counter() - tE
+                    if time_limit_s and \
+                            time.perf_counter() - t0 > time_limit_s:
+                        break
+                continue
             self._packed_eval_phase(evalmat)
             self.comm.all_reduce_sum_(evalmat)
             ev = evalmat.cpu().numpy()          # the round's one host sync
@@ -702,6 +810,12 @@ class DistributedRBCDDriver:
             if time_limit_s and time.perf_counter() - t0 > time_limit_s:
                 break
             selected = int(np.argmax(gn2))
+        if chunk > 1:
+            flush_ring()   # safety net: all exits above should have flushed
+        if _timing and self.comm.rank == 0 and res.iterations:
+            k = res.iterations
+            print("phase ms/round: solve_enq %.3f finish %.3f pack %.3f "
+                  "eval %.3f flush %.3f" % tuple(p / k * 1e3 for p in ph))
         if res.trace:
             res.final_cost, res.final_gradnorm = res.trace[-1]
         if fout:

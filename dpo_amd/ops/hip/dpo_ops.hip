@@ -1512,6 +1512,9 @@ struct DpoCtx {
   double* pend_X = nullptr;
   // data-flow fence slots (see k_fence_* above)
   unsigned int* fences = nullptr;
+  // fixed 3-double eval destination for the group fan-out path (a
+  // stable pointer keeps the eval graph cache hot)
+  double* eval3 = nullptr;
   void invalidate_graphs() {
     if (solve_graph) { hipGraphExecDestroy(solve_graph); solve_graph = nullptr; }
     if (eval_graph) { hipGraphExecDestroy(eval_graph); eval_graph = nullptr; }
@@ -1644,6 +1647,7 @@ void* dpo_ctx_create(int n, int d, int r, int max_inner) {
                                     hipEventDisableTiming));
   DPO_CHECK(hipMalloc(&c->fences, F_NUM * sizeof(unsigned int)));
   DPO_CHECK(hipMemset(c->fences, 0, F_NUM * sizeof(unsigned int)));
+  DPO_CHECK(hipMalloc(&c->eval3, 3 * sizeof(double)));
   return c;
 }
 
@@ -1658,7 +1662,7 @@ void dpo_ctx_destroy(void* h) {
   if (c->start_event) hipEventDestroy(c->start_event);
   if (c->done_event) hipEventDestroy(c->done_event);
   hipFree(c->ctrl); hipFree(c->G_buf); hipHostFree(c->ctrl_host);
-  hipFree(c->fences);
+  hipFree(c->fences); hipFree(c->eval3);
   delete c;
 }
 
@@ -2078,6 +2082,101 @@ void dpo_round_eval(void* h, const double* X, const double* nbr,
   hipStream_t s = (hipStream_t)stream;
   fence_signal(c, F_EVAL_IN, s);
   round_eval_impl(c, X, nbr, out_dev, s);
+}
+
+// --- multi-agent round fan-out --------------------------------------
+// The distributed driver runs one rank per GPU; within a rank several
+// agents share the device. A round's per-agent enqueue loops (solve
+// fan-out over the active color, eval fan-out over every agent) are
+// pure launch overhead when driven from Python (~35 us/agent of
+// interpreter + dispatcher time); DpoGroup runs each loop as ONE C
+// call and gathers the per-agent eval scalars with a single kernel.
+
+__global__ void k_gather3(double* __restrict__ out, long stride,
+                          double* const* __restrict__ srcs,
+                          const int* __restrict__ rows, int n) {
+  int i = blockIdx.x;
+  int j = threadIdx.x;
+  if (i < n && j < 3) out[(long)rows[i] * stride + j] = srcs[i][j];
+}
+
+struct DpoGroup {
+  int n;
+  DpoCtx** cs;
+  double** Xs;          // host arrays of fixed device pointers
+  const double** nbrs;
+  double** d_srcs;      // device array: each ctx's eval3
+  int* d_rows;          // device array: output row per agent
+};
+
+void dpo_group_destroy(void* g) {
+  DpoGroup* gr = (DpoGroup*)g;
+  if (!gr) return;
+  hipFree(gr->d_srcs);
+  hipFree(gr->d_rows);
+  free(gr->cs); free(gr->Xs); free(gr->nbrs);
+  delete gr;
+}
+
+void* dpo_group_create(void** handles, int n, double** Xs,
+                       const double** nbrs, const int* rows) {
+  if (n <= 0 || n > 256) return nullptr;
+  DpoGroup* gr = new DpoGroup();
+  gr->n = n;
+  gr->cs = (DpoCtx**)malloc(n * sizeof(void*));
+  gr->Xs = (double**)malloc(n * sizeof(void*));
+  gr->nbrs = (const double**)malloc(n * sizeof(void*));
+  double* srcs_h[256];
+  for (int i = 0; i < n; ++i) {
+    gr->cs[i] = (DpoCtx*)handles[i];
+    gr->Xs[i] = Xs[i];
+    gr->nbrs[i] = nbrs[i];
+    srcs_h[i] = gr->cs[i]->eval3;
+  }
+  DPO_CHECK(hipMalloc(&gr->d_srcs, n * sizeof(double*)));
+  DPO_CHECK(hipMemcpy(gr->d_srcs, srcs_h, n * sizeof(double*),
+                      hipMemcpyHostToDevice));
+  DPO_CHECK(hipMalloc(&gr->d_rows, n * sizeof(int)));
+  DPO_CHECK(hipMemcpy(gr->d_rows, rows, n * sizeof(int),
+                      hipMemcpyHostToDevice));
+  return gr;
+}
+
+// Fan-out solves for the active agents (indices into the group), on
+// each agent's private exec stream, fenced against join_stream.
+void dpo_group_solve_start(void* g, const int* ids, int k, double tol,
+                           double Delta0, double accept_rho,
+                           void* join_stream) {
+  DpoGroup* gr = (DpoGroup*)g;
+  for (int i = 0; i < k; ++i) {
+    int a = ids[i];
+    dpo_round_solve_async(gr->cs[a], gr->Xs[a], gr->nbrs[a], tol, Delta0,
+                          accept_rho, join_stream);
+  }
+}
+
+void dpo_group_solve_finish(void* g, const int* ids, int k,
+                            int max_shrink, void* join_stream) {
+  DpoGroup* gr = (DpoGroup*)g;
+  for (int i = 0; i < k; ++i)
+    dpo_round_solve_finish(gr->cs[ids[i]], max_shrink, nullptr,
+                           join_stream);
+}
+
+// Fan-out evaluation of EVERY agent in the group into its eval3
+// scratch (per-agent exec streams, fenced), then gather all rows into
+// out_dev[rows[i]*row_stride .. +3] with one kernel on join_stream.
+void dpo_group_eval(void* g, double* out_dev, long row_stride,
+                    void* join_stream) {
+  DpoGroup* gr = (DpoGroup*)g;
+  hipStream_t js = (hipStream_t)join_stream;
+  for (int i = 0; i < gr->n; ++i)
+    dpo_round_eval_async(gr->cs[i], gr->Xs[i], gr->nbrs[i],
+                         gr->cs[i]->eval3, join_stream);
+  for (int i = 0; i < gr->n; ++i)
+    DPO_CHECK(hipStreamWaitEvent(js, gr->cs[i]->done_event, 0));
+  hipLaunchKernelGGL(k_gather3, dim3(gr->n), dim3(64), 0, js,
+                     out_dev, row_stride, gr->d_srcs, gr->d_rows, gr->n);
 }
 
 }  // extern "C"

@@ -70,6 +70,16 @@ _lib.dpo_round_eval_async.argtypes = [_c, _c, _c, _c, _c]
 _lib.dpo_eval_join.argtypes = [_c, _c]
 _lib.dpo_gnc_weights.argtypes = [_c, _c, _c, _c, _c, _c, _c, _c, _c, _c,
                                  _c, _c, _c, _i, _i, _i, _d, _d, _c]
+_lib.dpo_group_create.restype = _c
+_lib.dpo_group_create.argtypes = [
+    ctypes.POINTER(ctypes.c_void_p), _i, ctypes.POINTER(ctypes.c_void_p),
+    ctypes.POINTER(ctypes.c_void_p), ctypes.POINTER(ctypes.c_int)]
+_lib.dpo_group_destroy.argtypes = [_c]
+_lib.dpo_group_solve_start.argtypes = [
+    _c, ctypes.POINTER(ctypes.c_int), _i, _d, _d, _d, _c]
+_lib.dpo_group_solve_finish.argtypes = [
+    _c, ctypes.POINTER(ctypes.c_int), _i, _i, _c]
+_lib.dpo_group_eval.argtypes = [_c, _c, _l, _c]
 
 CTRL_SIZE = _lib.dpo_ctrl_size()
 
@@ -324,6 +334,56 @@ class DeviceSolver:
     def eval_join(self, X: Tensor) -> Tensor:
         _lib.dpo_eval_join(self.handle, _stream(X))
         return self._eval_out
+
+
+class DeviceGroup:
+    """Multi-agent round fan-out: one C call per phase instead of a
+    Python loop over agents (solve fan-out for the active set, eval
+    fan-out + single-gather-kernel for every agent). Pointers (X,
+    neighbor buffer, eval scratch) are fixed at construction; the
+    per-ctx hipGraph caches stay hot."""
+
+    def __init__(self, solvers, Xs, nbrs, rows):
+        n = len(solvers)
+        assert n == len(Xs) == len(nbrs) == len(rows)
+        self._keep = (solvers, Xs, nbrs)
+        handles = (ctypes.c_void_p * n)(
+            *[ctypes.c_void_p(s.handle) for s in solvers])
+        xp = (ctypes.c_void_p * n)(*[ctypes.c_void_p(x.data_ptr())
+                                     for x in Xs])
+        np_ = (ctypes.c_void_p * n)(*[ctypes.c_void_p(b.data_ptr())
+                                      for b in nbrs])
+        rw = (ctypes.c_int * n)(*rows)
+        self.n = n
+        self.handle = _lib.dpo_group_create(handles, n, xp, np_, rw)
+        assert self.handle, "dpo_group_create failed"
+        self._js_of = Xs[0]
+
+    def __del__(self):
+        try:
+            if getattr(self, "handle", None):
+                _lib.dpo_group_destroy(self.handle)
+        except Exception:
+            pass
+
+    def ids(self, idx_list):
+        return (ctypes.c_int * len(idx_list))(*idx_list)
+
+    def solve_start(self, ids, tol: float = 1e-2, Delta0: float = 100.0,
+                    accept_rho: float = 0.1) -> None:
+        _lib.dpo_group_solve_start(self.handle, ids, len(ids), tol,
+                                   Delta0, accept_rho,
+                                   _stream(self._js_of))
+
+    def solve_finish(self, ids, max_shrink: int = 10) -> None:
+        _lib.dpo_group_solve_finish(self.handle, ids, len(ids),
+                                    max_shrink, _stream(self._js_of))
+
+    def eval_all(self, out: Tensor) -> None:
+        """Evaluate every agent; out is a (num_robots, 3) fp64 device
+        matrix — only this group's rows are written."""
+        _lib.dpo_group_eval(self.handle, _p(out), out.stride(0),
+                            _stream(self._js_of))
 
 
 def gnc_weights(X: Tensor, nbr: Tensor, g: dict, weights: Tensor,
