@@ -418,7 +418,7 @@ __global__ void k_proj_dots(const double* __restrict__ X,
 // MODE 1 (no projection, no store): dots <QV, V> and <G, V> (the
 // trust-region candidate's f evaluation).
 // ---------------------------------------------------------------------
-template <int D, int R, int MODE>
+template <int D, int R, int MODE, int CF = CF_NONE>
 __global__ void k_hess_fused(const int* __restrict__ row_ptr,
                              const int* __restrict__ col_idx,
                              const double* __restrict__ vals,
@@ -527,6 +527,10 @@ __global__ void k_hess_fused(const int* __restrict__ row_ptr,
   }
   if (dot_slot >= 0) block_reduce_atomic(d0, ctrl + dot_slot);
   if (dot_slot2 >= 0) block_reduce_atomic(d1, ctrl + dot_slot2);
+  if (CF != CF_NONE) {
+    if (fanin_last_block(ctrl) && threadIdx.x == 0)
+      run_ctrl_tail(CF, ctrl, nullptr);
+  }
 }
 
 // ---------------------------------------------------------------------
@@ -773,7 +777,9 @@ __global__ void k_tcg_update(double* __restrict__ eta,
   double rr = 0.0;
   if (i < total) {
     const double dl = delta[i];
-    const double et = eta[i];
+    // j == 0: eta starts the solve at 0 — write instead of accumulate,
+    // so the eta buffer needs no per-solve zeroing pass
+    const double et = (j == 0) ? 0.0 : eta[i];
     eta_snap[(size_t)j * total + i] = et;
     delta_snap[(size_t)j * total + i] = dl;
     eta[i] = fma(coef, dl, et);
@@ -1219,7 +1225,7 @@ static void launch_polar(const double* A, const double* B, const double* C,
   dpo_bad_shape(d, r);
 }
 
-template <int MODE>
+template <int MODE, int CF = CF_NONE>
 static void launch_hess_fused(const int* rp, const int* ci,
                               const double* vals, const double* V,
                               const double* X, const double* G,
@@ -1230,14 +1236,29 @@ static void launch_hess_fused(const int* rp, const int* ci,
   const int grid = blocks_for(n, 256);
 #define CASE_HF(D, R) \
   if (d == D && r == R) { \
-    hipLaunchKernelGGL((k_hess_fused<D, R, MODE>), dim3(grid), dim3(256), \
-                       0, s, rp, ci, vals, V, X, G, out, dotW, ctrl, n, \
-                       dot_slot, dot_slot2, guard); \
+    hipLaunchKernelGGL((k_hess_fused<D, R, MODE, CF>), dim3(grid), \
+                       dim3(256), 0, s, rp, ci, vals, V, X, G, out, dotW, \
+                       ctrl, n, dot_slot, dot_slot2, guard); \
     return; \
   }
   DPO_FOREACH_DR(CASE_HF)
 #undef CASE_HF
   dpo_bad_shape(d, r);
+}
+
+// Graph-safe zero / device->pinned-host copy. hipMemsetAsync and
+// hipMemcpyAsync become SDMA-engine nodes inside a captured hipGraph;
+// compute<->SDMA dependencies in replayed graphs are part of the same
+// ROCm 7.2 ordering fragility the launch fences work around, so the
+// captured bodies use plain kernels only.
+__global__ void k_dzero(double* p, long n) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) p[i] = 0.0;
+}
+__global__ void k_ctrl_to_host(const double* __restrict__ src,
+                               double* __restrict__ dst, int n) {
+  int i = threadIdx.x;
+  if (i < n) dst[i] = src[i];
 }
 
 static void launch_precond_dense(const float* Minv, const double* V,
@@ -1248,8 +1269,12 @@ static void launch_precond_dense(const float* Minv, const double* V,
   // j-split so small problems still produce >= ~512 workgroups
   int jsplit = 1;
   while (gx * jsplit < 512 && jsplit < 32) jsplit *= 2;
-  if (jsplit > 1)
-    DPO_CHECK(hipMemsetAsync(Z, 0, (size_t)N * r * sizeof(double), s));
+  if (jsplit > 1) {
+    // kernel (not hipMemsetAsync): this runs inside captured graphs,
+    // which must stay free of SDMA nodes (see fence comment)
+    hipLaunchKernelGGL(k_dzero, dim3(blocks_for((long)N * r, 256)),
+                       dim3(256), 0, s, Z, (long)N * r);
+  }
 #define CASE_PD(D, R) \
   if (r == R) { \
     hipLaunchKernelGGL((k_precond_dense<R>), dim3(gx, jsplit), dim3(256), \
@@ -1569,20 +1594,6 @@ static inline void fence_wait(DpoCtx* c, int which, hipStream_t s) {
                      c->fences + which);
 }
 
-// Graph-safe zero / device->pinned-host copy. hipMemsetAsync and
-// hipMemcpyAsync become SDMA-engine nodes inside a captured hipGraph;
-// compute<->SDMA dependencies in replayed graphs are part of the same
-// ROCm 7.2 ordering fragility the fences work around, so the captured
-// bodies use plain kernels only.
-__global__ void k_dzero(double* p, long n) {
-  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i < n) p[i] = 0.0;
-}
-__global__ void k_ctrl_to_host(const double* __restrict__ src,
-                               double* __restrict__ dst, int n) {
-  int i = threadIdx.x;
-  if (i < n) dst[i] = src[i];
-}
 static inline void dzero(double* p, long n, hipStream_t s) {
   hipLaunchKernelGGL(k_dzero, dim3((unsigned)((n + 255) / 256)), dim3(256),
                      0, s, p, n);
@@ -1634,6 +1645,12 @@ void* dpo_ctx_create(int n, int d, int r, int max_inner) {
   DPO_CHECK(hipMalloc(&c->delta_snap, vb * (max_inner + 1)));
   DPO_CHECK(hipMalloc(&c->ctrl, CTRL_SIZE * sizeof(double)));
   DPO_CHECK(hipMalloc(&c->G_buf, vb));
+  // delta is consumed as `beta*delta - z` with beta == 0 on the first
+  // tCG iteration: one-time zeroing keeps 0*garbage (possible NaN bit
+  // patterns in fresh allocations) out of the recurrence, and the
+  // per-solve zeroing passes are dropped from the captured body
+  DPO_CHECK(hipMemset(c->delta, 0, vb));
+  DPO_CHECK(hipMemset(c->eta, 0, vb));
   DPO_CHECK(hipHostMalloc(&c->ctrl_host, CTRL_SIZE * sizeof(double)));
   DPO_CHECK(hipHostGetDevicePointer((void**)&c->ctrl_host_dev,
                                     c->ctrl_host, 0));
@@ -1686,11 +1703,19 @@ static void enqueue_solve_body(DpoCtx* c, double* X, const double* nbr,
   const int n = c->n, d = c->d, r = c->r;
   const long total = c->total;
   const int gvec = (int)((total + 255) / 256);
+  // Fused control tails (CF_*): the tCG scalar decisions run in the
+  // last-arriving block of the producing reduction kernel instead of a
+  // dedicated single-wave kernel, removing 3 of 8 launches per tCG
+  // iteration. The ~4 us/launch dispatch latency floor is what bounds
+  // these small-agent solves (see profiles/), so launch count is the
+  // lever. An earlier attempt at this fusion showed trajectory scatter
+  // that was eventually root-caused to an LDS reuse race in
+  // block_reduce_atomic plus the hipGraph ordering bugs (both fixed);
+  // DPO_NO_CF=1 restores the dedicated control kernels.
+  static const bool no_cf = dpo_env_flag("DPO_NO_CF");
   fence_wait(c, F_SOLVE_IN, s);  // first node: block until inputs ready
   if (nbr) ctx_assemble_g(c, nbr, s);
   dzero(c->ctrl, CTRL_SIZE, s);
-  dzero(c->eta, total, s);
-  dzero(c->delta, total, s);
 
   // gradient phase
   ctx_spmm(c, X, c->W, -1, s);
@@ -1707,32 +1732,45 @@ static void enqueue_solve_body(DpoCtx* c, double* X, const double* nbr,
                      Delta0, 1.0, 0.1);
   // z0
   ctx_precond(c, c->rvec, c->z, s);
-  launch_proj_dots(X, c->z, nullptr, c->z, c->rvec, c->ctrl, n, d, r,
-                   C_DOT0, -1, ST_RUN, s);
-  hipLaunchKernelGGL(k_ctrl_z0, dim3(1), dim3(64), 0, s, c->ctrl);
+  if (no_cf) {
+    launch_proj_dots(X, c->z, nullptr, c->z, c->rvec, c->ctrl, n, d, r,
+                     C_DOT0, -1, ST_RUN, s);
+    hipLaunchKernelGGL(k_ctrl_z0, dim3(1), dim3(64), 0, s, c->ctrl);
+  } else {
+    launch_proj_dots_cf<CF_Z0>(X, c->z, nullptr, c->z, c->rvec, c->ctrl,
+                               n, d, r, C_DOT0, -1, ST_RUN, s);
+  }
   hipLaunchKernelGGL(k_tcg_delta, dim3(gvec), dim3(256), 0, s,
                      c->delta, c->z, c->ctrl, total);
 
-  // tCG loop. NOTE: a fused fan-in variant (control logic as split-K
-  // tails of the reduction kernels, CF_* templates above) measured +8%
-  // throughput but showed intermittent run-to-run trajectory scatter on
-  // full convergence runs; until that is root-caused the control
-  // decisions run as dedicated single-wave kernels (each kernel
-  // boundary is an agent-scope acquire, making the cross-workgroup
-  // hand-off trivially correct).
   for (int j = 0; j < c->max_inner; ++j) {
-    launch_hess_fused<0>(c->q_rp, c->q_ci, c->q_vals, c->delta, X,
-                         nullptr, c->Hd, c->delta, c->ctrl, n, d, r,
-                         C_DOT0, -1, ST_RUN, s);
-    hipLaunchKernelGGL(k_ctrl_alpha, dim3(1), dim3(64), 0, s, c->ctrl);
-    hipLaunchKernelGGL((k_tcg_update<CF_NONE>), dim3(gvec), dim3(256), 0,
-                       s, c->eta, c->rvec, c->delta, c->Hd, c->eta_snap,
-                       c->delta_snap, c->ctrl, total);
-    hipLaunchKernelGGL(k_ctrl_rr, dim3(1), dim3(64), 0, s, c->ctrl);
+    if (no_cf) {
+      launch_hess_fused<0>(c->q_rp, c->q_ci, c->q_vals, c->delta, X,
+                           nullptr, c->Hd, c->delta, c->ctrl, n, d, r,
+                           C_DOT0, -1, ST_RUN, s);
+      hipLaunchKernelGGL(k_ctrl_alpha, dim3(1), dim3(64), 0, s, c->ctrl);
+      hipLaunchKernelGGL((k_tcg_update<CF_NONE>), dim3(gvec), dim3(256),
+                         0, s, c->eta, c->rvec, c->delta, c->Hd,
+                         c->eta_snap, c->delta_snap, c->ctrl, total);
+      hipLaunchKernelGGL(k_ctrl_rr, dim3(1), dim3(64), 0, s, c->ctrl);
+    } else {
+      launch_hess_fused<0, CF_ALPHA>(
+          c->q_rp, c->q_ci, c->q_vals, c->delta, X, nullptr, c->Hd,
+          c->delta, c->ctrl, n, d, r, C_DOT0, -1, ST_RUN, s);
+      hipLaunchKernelGGL((k_tcg_update<CF_RR>), dim3(gvec), dim3(256),
+                         0, s, c->eta, c->rvec, c->delta, c->Hd,
+                         c->eta_snap, c->delta_snap, c->ctrl, total);
+    }
     ctx_precond(c, c->rvec, c->z, s);
-    launch_proj_dots(X, c->z, nullptr, c->z, c->rvec, c->ctrl, n, d, r,
-                     C_DOT0, -1, ST_RUN, s);
-    hipLaunchKernelGGL(k_ctrl_beta, dim3(1), dim3(64), 0, s, c->ctrl);
+    if (no_cf) {
+      launch_proj_dots(X, c->z, nullptr, c->z, c->rvec, c->ctrl, n, d, r,
+                       C_DOT0, -1, ST_RUN, s);
+      hipLaunchKernelGGL(k_ctrl_beta, dim3(1), dim3(64), 0, s, c->ctrl);
+    } else {
+      launch_proj_dots_cf<CF_BETA>(X, c->z, nullptr, c->z, c->rvec,
+                                   c->ctrl, n, d, r, C_DOT0, -1, ST_RUN,
+                                   s);
+    }
     hipLaunchKernelGGL(k_tcg_delta, dim3(gvec), dim3(256), 0, s,
                        c->delta, c->z, c->ctrl, total);
   }
@@ -1973,10 +2011,19 @@ void dpo_eval_terms(void* h, const double* X, double* out_dev,
   const int n = c->n, d = c->d, r = c->r;
   const long total = c->total;
   const int gvec = (int)((total + 255) / 256);
+  static const bool no_cf = dpo_env_flag("DPO_NO_CF");
   dzero(c->ctrl + C_DOT0, 4, s);
   ctx_spmm(c, X, c->W, -1, s);
   launch_proj_dots(X, c->W, c->Gt, c->grad, nullptr, c->ctrl, n,
                    d, r, C_DOT1, C_DOT0, -1, s);
+  if (c->Gt && !no_cf) {
+    // the <G, X> dot's last block combines all three scalars into
+    // out_dev (CF_COMBINE tail) instead of a separate kernel
+    hipLaunchKernelGGL((k_dots<CF_COMBINE>), dim3(gvec), dim3(256), 0, s,
+                       c->Gt, X, (const double*)nullptr, c->ctrl,
+                       C_DOT2, -1, total, -1, out_dev);
+    return;
+  }
   if (c->Gt)
     hipLaunchKernelGGL(k_dots, dim3(gvec), dim3(256), 0, s,
                        c->Gt, X, (const double*)nullptr, c->ctrl,
