@@ -862,6 +862,235 @@ __global__ void k_dots(const double* __restrict__ A,
 }
 
 // ---------------------------------------------------------------------
+// Persistent tCG loop: the whole inner Steihaug iteration as ONE kernel
+// with device-side grid barriers between stages, instead of ~5 dependent
+// launches per iteration at the ~4 us dispatch-latency floor. Valid when
+// every stage fits one resident grid (blocks <= 256 on 256 CUs => all
+// workgroups co-resident, so the barrier cannot deadlock) and the dense
+// preconditioner is in use. Reduction tails reuse the CF fan-in
+// machinery; stage ordering/visibility comes from a generation-counter
+// barrier with agent-scope fences.
+// ---------------------------------------------------------------------
+__device__ __forceinline__ void grid_barrier(unsigned int* cnt,
+                                             unsigned int* gen,
+                                             unsigned int nb) {
+  __syncthreads();
+  __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+  if (threadIdx.x == 0) {
+    const unsigned int g =
+        __hip_atomic_load(gen, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    const unsigned int old = __hip_atomic_fetch_add(
+        cnt, 1u, __ATOMIC_ACQ_REL, __HIP_MEMORY_SCOPE_AGENT);
+    if (old == nb - 1) {
+      __hip_atomic_store(cnt, 0u, __ATOMIC_RELAXED,
+                         __HIP_MEMORY_SCOPE_AGENT);
+      __hip_atomic_fetch_add(gen, 1u, __ATOMIC_RELEASE,
+                             __HIP_MEMORY_SCOPE_AGENT);
+    } else {
+      long spins = 0;
+      while (__hip_atomic_load(gen, __ATOMIC_ACQUIRE,
+                               __HIP_MEMORY_SCOPE_AGENT) == g) {
+        __builtin_amdgcn_s_sleep(8);
+        if (++spins > 400000000L) __builtin_trap();  // lost-block guard
+      }
+    }
+  }
+  __syncthreads();
+  __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+}
+
+template <int D, int R>
+__global__ void k_tcg_persist(const int* __restrict__ q_rp,
+                              const int* __restrict__ q_ci,
+                              const double* __restrict__ q_vals,
+                              const double* __restrict__ X,
+                              const float* __restrict__ Minv,
+                              double* __restrict__ eta,
+                              double* __restrict__ rvec,
+                              double* __restrict__ delta,
+                              double* __restrict__ z,
+                              double* __restrict__ Hd,
+                              double* __restrict__ eta_snap,
+                              double* __restrict__ delta_snap,
+                              double* __restrict__ ctrl,
+                              unsigned int* __restrict__ gbar,
+                              int n, int N, long total, int max_inner) {
+  constexpr int dh = D + 1;
+  const int tid = blockIdx.x * blockDim.x + threadIdx.x;
+  const unsigned int nb = gridDim.x;
+  unsigned int* cnt = gbar;
+  unsigned int* gen = gbar + 1;
+  for (int it = 0; it < max_inner; ++it) {
+    grid_barrier(cnt, gen, nb);  // prior delta writes visible
+    if (ctrl[C_STATUS] != (double)ST_RUN) break;
+
+    // --- stage H: Hd = P_X(Q delta), d_Qd -> C_DOT0, tail_alpha -----
+    {
+      double d0 = 0.0;
+      if (tid < n) {
+        double acc[dh][R];
+        #pragma unroll
+        for (int c = 0; c < dh; ++c)
+          #pragma unroll
+          for (int k = 0; k < R; ++k) acc[c][k] = 0.0;
+        const int s0 = q_rp[tid], e0 = q_rp[tid + 1];
+        for (int p = s0; p < e0; ++p) {
+          const double* B = q_vals + (size_t)p * dh * dh;
+          const double* Vj = delta + (size_t)q_ci[p] * dh * R;
+          #pragma unroll
+          for (int c = 0; c < dh; ++c)
+            #pragma unroll
+            for (int cc = 0; cc < dh; ++cc) {
+              const double b = B[c * dh + cc];
+              #pragma unroll
+              for (int k = 0; k < R; ++k)
+                acc[c][k] = fma(b, Vj[cc * R + k], acc[c][k]);
+            }
+        }
+        const double* Xi = X + (size_t)tid * dh * R;
+        double S[D][D];
+        #pragma unroll
+        for (int a = 0; a < D; ++a)
+          #pragma unroll
+          for (int b = 0; b < D; ++b) {
+            double sv = 0.0;
+            #pragma unroll
+            for (int k = 0; k < R; ++k)
+              sv = fma(Xi[a * R + k], acc[b][k], sv);
+            S[a][b] = sv;
+          }
+        #pragma unroll
+        for (int a = 0; a < D; ++a)
+          #pragma unroll
+          for (int b = a; b < D; ++b) {
+            const double sv = 0.5 * (S[a][b] + S[b][a]);
+            S[a][b] = sv;
+            S[b][a] = sv;
+          }
+        #pragma unroll
+        for (int a = 0; a < D; ++a)
+          #pragma unroll
+          for (int k = 0; k < R; ++k) {
+            double v = acc[a][k];
+            #pragma unroll
+            for (int b = 0; b < D; ++b)
+              v = fma(-S[a][b], Xi[b * R + k], v);
+            acc[a][k] = v;
+          }
+        double* Oi = Hd + (size_t)tid * dh * R;
+        const double* Wi = delta + (size_t)tid * dh * R;
+        #pragma unroll
+        for (int c = 0; c < dh; ++c)
+          #pragma unroll
+          for (int k = 0; k < R; ++k) {
+            Oi[c * R + k] = acc[c][k];
+            d0 = fma(acc[c][k], Wi[c * R + k], d0);
+          }
+      }
+      block_reduce_atomic(d0, ctrl + C_DOT0);
+      if (fanin_last_block(ctrl) && threadIdx.x == 0)
+        ctrl_tail_alpha(ctrl);
+    }
+    grid_barrier(cnt, gen, nb);
+
+    // --- stage U: eta/r update + snapshots, rr -> C_DOT1, tail_rr ---
+    {
+      const double coef = ctrl[C_COEF];
+      const int stop_pending = (int)ctrl[C_STOP_PENDING];
+      const int j = (int)ctrl[C_ITER];
+      double rr = 0.0;
+      if (tid < total) {
+        const double dl = delta[tid];
+        const double et = (j == 0) ? 0.0 : eta[tid];
+        eta_snap[(size_t)j * total + tid] = et;
+        delta_snap[(size_t)j * total + tid] = dl;
+        eta[tid] = fma(coef, dl, et);
+        if (!stop_pending) {
+          const double rn = fma(coef, Hd[tid], rvec[tid]);
+          rvec[tid] = rn;
+          rr = rn * rn;
+        }
+      }
+      if (!stop_pending) block_reduce_atomic(rr, ctrl + C_DOT1);
+      if (fanin_last_block(ctrl) && threadIdx.x == 0)
+        ctrl_tail_rr(ctrl);
+    }
+    grid_barrier(cnt, gen, nb);
+    if (ctrl[C_STATUS] != (double)ST_RUN) break;
+
+    // --- stage P: z = Minv r (dense fp32 apply, one element/thread) -
+    if (tid < total) {
+      const int i = tid / R, k = tid % R;
+      double acc = 0.0;
+      for (int jj = 0; jj < N; ++jj)
+        acc = fma((double)Minv[(size_t)jj * N + i], rvec[(size_t)jj * R + k],
+                  acc);
+      z[tid] = acc;
+    }
+    grid_barrier(cnt, gen, nb);
+
+    // --- stage J: project z at X, <z, r> -> C_DOT0, tail_beta -------
+    {
+      double d0 = 0.0;
+      if (tid < n) {
+        const double* Xi = X + (size_t)tid * dh * R;
+        double* Zi = z + (size_t)tid * dh * R;
+        const double* Ri = rvec + (size_t)tid * dh * R;
+        double Vt[dh][R];
+        #pragma unroll
+        for (int c = 0; c < dh; ++c)
+          #pragma unroll
+          for (int k = 0; k < R; ++k) Vt[c][k] = Zi[c * R + k];
+        double S[D][D];
+        #pragma unroll
+        for (int a = 0; a < D; ++a)
+          #pragma unroll
+          for (int b = 0; b < D; ++b) {
+            double sv = 0.0;
+            #pragma unroll
+            for (int k = 0; k < R; ++k)
+              sv = fma(Xi[a * R + k], Vt[b][k], sv);
+            S[a][b] = sv;
+          }
+        #pragma unroll
+        for (int a = 0; a < D; ++a)
+          #pragma unroll
+          for (int b = a; b < D; ++b) {
+            const double sv = 0.5 * (S[a][b] + S[b][a]);
+            S[a][b] = sv;
+            S[b][a] = sv;
+          }
+        #pragma unroll
+        for (int a = 0; a < D; ++a)
+          #pragma unroll
+          for (int k = 0; k < R; ++k) {
+            double v = Vt[a][k];
+            #pragma unroll
+            for (int b = 0; b < D; ++b)
+              v = fma(-S[a][b], Xi[b * R + k], v);
+            Vt[a][k] = v;
+          }
+        #pragma unroll
+        for (int c = 0; c < dh; ++c)
+          #pragma unroll
+          for (int k = 0; k < R; ++k) {
+            Zi[c * R + k] = Vt[c][k];
+            d0 = fma(Vt[c][k], Ri[c * R + k], d0);
+          }
+      }
+      block_reduce_atomic(d0, ctrl + C_DOT0);
+      if (fanin_last_block(ctrl) && threadIdx.x == 0)
+        ctrl_tail_beta(ctrl);
+    }
+    grid_barrier(cnt, gen, nb);
+
+    // --- stage D: delta = beta*delta - z ----------------------------
+    if (tid < total)
+      delta[tid] = fma(ctrl[C_BETA], delta[tid], -z[tid]);
+  }
+}
+
+// ---------------------------------------------------------------------
 // Single-wave tCG control kernels (device-side branch logic)
 // ---------------------------------------------------------------------
 __global__ void k_ctrl_init(double* ctrl, double tol, double Delta0,
@@ -1537,6 +1766,8 @@ struct DpoCtx {
   double* pend_X = nullptr;
   // data-flow fence slots (see k_fence_* above)
   unsigned int* fences = nullptr;
+  // grid-barrier state for the persistent tCG kernel: [count, generation]
+  unsigned int* gbar = nullptr;
   // fixed 3-double eval destination for the group fan-out path (a
   // stable pointer keeps the eval graph cache hot)
   double* eval3 = nullptr;
@@ -1665,6 +1896,8 @@ void* dpo_ctx_create(int n, int d, int r, int max_inner) {
   DPO_CHECK(hipMalloc(&c->fences, F_NUM * sizeof(unsigned int)));
   DPO_CHECK(hipMemset(c->fences, 0, F_NUM * sizeof(unsigned int)));
   DPO_CHECK(hipMalloc(&c->eval3, 3 * sizeof(double)));
+  DPO_CHECK(hipMalloc(&c->gbar, 2 * sizeof(unsigned int)));
+  DPO_CHECK(hipMemset(c->gbar, 0, 2 * sizeof(unsigned int)));
   return c;
 }
 
@@ -1679,7 +1912,7 @@ void dpo_ctx_destroy(void* h) {
   if (c->start_event) hipEventDestroy(c->start_event);
   if (c->done_event) hipEventDestroy(c->done_event);
   hipFree(c->ctrl); hipFree(c->G_buf); hipHostFree(c->ctrl_host);
-  hipFree(c->fences); hipFree(c->eval3);
+  hipFree(c->fences); hipFree(c->eval3); hipFree(c->gbar);
   delete c;
 }
 
@@ -1692,6 +1925,22 @@ void dpo_ctx_set_problem(void* h, const int* rp, const int* ci,
     c->invalidate_graphs();
   c->q_rp = rp; c->q_ci = ci; c->q_vals = vals;
   c->Gt = Gt; c->Minv = Minv; c->Ljac = Ljac;
+}
+
+static void launch_tcg_persist(DpoCtx* c, const double* X, hipStream_t s) {
+  const int gvec = (int)((c->total + 255) / 256);
+#define CASE_TP(D, R) \
+  if (c->d == D && c->r == R) { \
+    hipLaunchKernelGGL((k_tcg_persist<D, R>), dim3(gvec), dim3(256), 0, s, \
+                       c->q_rp, c->q_ci, c->q_vals, X, c->Minv, c->eta, \
+                       c->rvec, c->delta, c->z, c->Hd, c->eta_snap, \
+                       c->delta_snap, c->ctrl, c->gbar, c->n, c->N, \
+                       c->total, c->max_inner); \
+    return; \
+  }
+  DPO_FOREACH_DR(CASE_TP)
+#undef CASE_TP
+  dpo_bad_shape(c->d, c->r);
 }
 
 // Enqueue the fixed pre-sync solve sequence (gradient, tCG, first
@@ -1743,6 +1992,17 @@ static void enqueue_solve_body(DpoCtx* c, double* X, const double* nbr,
   hipLaunchKernelGGL(k_tcg_delta, dim3(gvec), dim3(256), 0, s,
                      c->delta, c->z, c->ctrl, total);
 
+  // Persistent whole-loop kernel when the grid is small enough for
+  // guaranteed co-residency (<= 256 workgroups on 256 CUs) and the
+  // dense preconditioner is bound; DPO_NO_PERSIST=1 falls back to the
+  // per-stage launch sequence below.
+  static const bool no_persist = dpo_env_flag("DPO_NO_PERSIST");
+  if (!no_persist && !no_cf && c->Minv != nullptr && gvec <= 256) {
+    launch_tcg_persist(c, X, s);
+    hipLaunchKernelGGL(k_ctrl_tcg_end, dim3(1), dim3(64), 0, s, c->ctrl);
+    goto candidate;
+  }
+
   for (int j = 0; j < c->max_inner; ++j) {
     if (no_cf) {
       launch_hess_fused<0>(c->q_rp, c->q_ci, c->q_vals, c->delta, X,
@@ -1775,6 +2035,7 @@ static void enqueue_solve_body(DpoCtx* c, double* X, const double* nbr,
                        c->delta, c->z, c->ctrl, total);
   }
   hipLaunchKernelGGL(k_ctrl_tcg_end, dim3(1), dim3(64), 0, s, c->ctrl);
+candidate:
   // first candidate attempt is part of the fixed sequence
   hipLaunchKernelGGL(k_ctrl_candidate, dim3(1), dim3(64), 0, s, c->ctrl);
   hipLaunchKernelGGL(k_form_step, dim3(gvec), dim3(256), 0, s,

@@ -212,3 +212,31 @@ class QuadraticOptimizer:
         problem = self.problem
         grad = problem.rie_grad(X0)
         return problem.manifold.retract(X0, -self.gd_stepsize * grad)
+
+    def gradient_descent_ls(self, X0: Tensor,
+                            max_iterations: int = 10) -> Tensor:
+        """Backtracking line-search steepest descent (the reference's
+        RSD variant, QuadraticOptimizer.cpp:151-172 — present in its
+        API surface though unused by the agents). Armijo condition with
+        step halving from a Riemannian-gradient-scaled initial step."""
+        problem = self.problem
+        X = X0
+        fX = problem.f(X)
+        for _ in range(max_iterations):
+            grad = problem.rie_grad(X)
+            gn2 = float((grad * grad).sum())
+            if gn2 == 0.0:
+                break
+            t = 1.0 / (1.0 + gn2 ** 0.5)
+            accepted = False
+            for _bt in range(20):
+                X_prop = problem.manifold.retract(X, -t * grad)
+                f_prop = problem.f(X_prop)
+                if f_prop <= fX - 1e-4 * t * gn2:
+                    X, fX = X_prop, f_prop
+                    accepted = True
+                    break
+                t *= 0.5
+            if not accepted:
+                break
+        return X

@@ -198,3 +198,23 @@ def test_checkpoint_resume(tmp_path):
     assert abs(f_resumed - f_before) < 1e-6 * max(1.0, abs(f_before))
     b.iterate(True)  # keeps optimizing from the checkpoint
     assert b.problem.f(b.X) <= f_resumed + 1e-9
+
+
+def test_gradient_descent_ls_descends():
+    # reference QuadraticOptimizer::gradientDescentLS (RSD line search)
+    meas, n = grid3d(side=2, seed=7, rot_noise=0.3, tran_noise=0.2)
+    d = 3
+    from dpo_amd.chordal import odometry_initialization
+    odo = [m for m in meas if m.p1 + 1 == m.p2]
+    T = odometry_initialization(d, n, odo)
+    Q = assemble_connection_laplacian(meas, n, d)
+    prob = QuadraticProblem(n, d, d)
+    prob.set_q(Q)
+    X = torch.from_numpy(np.ascontiguousarray(T.T))
+    opt = QuadraticOptimizer(prob, OptAlgorithm.RGD, TRParams())
+    f0 = prob.f(X)
+    Xn = opt.gradient_descent_ls(X, max_iterations=8)
+    f1 = prob.f(Xn)
+    assert f1 <= f0
+    # line search should make real progress on a noisy instance
+    assert f1 < f0 - 1e-6 * max(1.0, abs(f0))
