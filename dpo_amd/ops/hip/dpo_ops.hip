@@ -862,14 +862,19 @@ __global__ void k_dots(const double* __restrict__ A,
 }
 
 // ---------------------------------------------------------------------
-// Persistent tCG loop: the whole inner Steihaug iteration as ONE kernel
-// with device-side grid barriers between stages, instead of ~5 dependent
-// launches per iteration at the ~4 us dispatch-latency floor. Valid when
-// every stage fits one resident grid (blocks <= 256 on 256 CUs => all
-// workgroups co-resident, so the barrier cannot deadlock) and the dense
-// preconditioner is in use. Reduction tails reuse the CF fan-in
-// machinery; stage ordering/visibility comes from a generation-counter
-// barrier with agent-scope fences.
+// Persistent solve kernel: the ENTIRE pre-sync local solve — G
+// assembly, gradient phase, z0, the Steihaug tCG loop, the first
+// candidate (replay + polar retraction + f evaluation) and the
+// acceptance test — as ONE kernel with device-side grid barriers
+// between stages, instead of ~60 dependent launches at the ~4 us
+// dispatch-latency floor. Valid when every stage fits one resident
+// grid (blocks <= 256 on 256 CUs => all workgroups co-resident, so
+// the barrier cannot deadlock) and the dense preconditioner is in
+// use. Reduction tails reuse the CF fan-in machinery; stage
+// ordering/visibility comes from a generation-counter barrier with
+// agent-scope fences. The control block is written to mapped pinned
+// host memory by the last stage, so the host-side acceptance /
+// shrink logic (solve_postsync) proceeds after one stream sync.
 // ---------------------------------------------------------------------
 __device__ __forceinline__ void grid_barrier(unsigned int* cnt,
                                              unsigned int* gen,
@@ -899,137 +904,239 @@ __device__ __forceinline__ void grid_barrier(unsigned int* cnt,
   __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
 }
 
+// k_ctrl_candidate's body as a device function (single thread).
+__device__ void ctrl_candidate_logic(double* ctrl) {
+  if (ctrl[C_STATUS] != (double)ST_TCG_STOP) return;
+  const double radius = ctrl[C_RADIUS];
+  const int hlen = (int)ctrl[C_HLEN];
+  double dm = 0.0;
+  int jstar = -1;
+  double taustar = 0.0;
+  for (int j = 0; j < hlen && j < MAX_TCG; ++j) {
+    const double z_r = ctrl[H_ZR(j)];
+    const double d_Hd = ctrl[H_DHD(j)];
+    const double e_Pe = ctrl[H_EPE(j)];
+    const double e_Pd = ctrl[H_EPD(j)];
+    const double d_Pd = ctrl[H_DPD(j)];
+    const double alpha = z_r / d_Hd;
+    const double e_Pe_new = e_Pe + 2.0 * alpha * e_Pd + alpha * alpha * d_Pd;
+    if (d_Hd <= 0.0 || e_Pe_new >= radius * radius) {
+      const double disc = e_Pd * e_Pd + d_Pd * (radius * radius - e_Pe);
+      const double tau = (d_Pd > 0.0)
+          ? (-e_Pd + sqrt(fmax(disc, 0.0))) / d_Pd : 0.0;
+      dm += tau * z_r - 0.5 * tau * tau * d_Hd;
+      jstar = j;
+      taustar = tau;
+      break;
+    }
+    dm += 0.5 * alpha * z_r;
+  }
+  if (jstar < 0) {
+    ctrl[C_USE_CURRENT] = 1.0;
+  } else {
+    ctrl[C_USE_CURRENT] = 0.0;
+    ctrl[C_JSTAR] = (double)jstar;
+    ctrl[C_TAUSTAR] = taustar;
+  }
+  ctrl[C_DM] = dm;
+  ctrl[C_DOT0] = 0.0;
+  ctrl[C_DOT2] = 0.0;
+}
+
 template <int D, int R>
-__global__ void k_tcg_persist(const int* __restrict__ q_rp,
-                              const int* __restrict__ q_ci,
-                              const double* __restrict__ q_vals,
-                              const double* __restrict__ X,
-                              const float* __restrict__ Minv,
-                              double* __restrict__ eta,
-                              double* __restrict__ rvec,
-                              double* __restrict__ delta,
-                              double* __restrict__ z,
-                              double* __restrict__ Hd,
-                              double* __restrict__ eta_snap,
-                              double* __restrict__ delta_snap,
-                              double* __restrict__ ctrl,
-                              unsigned int* __restrict__ gbar,
-                              int n, int N, long total, int max_inner) {
+__global__ void k_solve_persist(
+    const int* __restrict__ q_rp, const int* __restrict__ q_ci,
+    const double* __restrict__ q_vals, const double* __restrict__ X,
+    const double* __restrict__ G,   // linear term (read; may be null)
+    double* __restrict__ Gw,        // == G when assembling here, else null
+    const double* __restrict__ gE0, const long* __restrict__ g_local_pose,
+    const long* __restrict__ g_nbr_slot, const double* __restrict__ nbr,
+    const double* __restrict__ g_w, int g_ne,
+    const float* __restrict__ Minv, double* __restrict__ eta,
+    double* __restrict__ rvec, double* __restrict__ delta,
+    double* __restrict__ z, double* __restrict__ Hd,
+    double* __restrict__ eta_snap, double* __restrict__ delta_snap,
+    double* __restrict__ step, double* __restrict__ Xprop,
+    double* __restrict__ ctrl, double* __restrict__ ctrl_host,
+    unsigned int* __restrict__ gbar, int n, int N, long total,
+    int max_inner, double tol, double Delta0, double theta, double kappa,
+    double accept_rho, int jmul, int full) {
   constexpr int dh = D + 1;
   const int tid = blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
   const unsigned int nb = gridDim.x;
   unsigned int* cnt = gbar;
   unsigned int* gen = gbar + 1;
-  for (int it = 0; it < max_inner; ++it) {
-    grid_barrier(cnt, gen, nb);  // prior delta writes visible
-    if (ctrl[C_STATUS] != (double)ST_RUN) break;
+#define PSTAMP(i) \
+  if (blockIdx.x == 0 && threadIdx.x == 0) \
+    ctrl_host[CTRL_SIZE + (i)] = (double)wall_clock64();
+  PSTAMP(0)
 
-    // --- stage H: Hd = P_X(Q delta), d_Qd -> C_DOT0, tail_alpha -----
-    {
-      double d0 = 0.0;
-      if (tid < n) {
-        double acc[dh][R];
+  if (full) {
+  // --- S0: zero ctrl (+ G when assembling) --------------------------
+  for (long t = tid; t < CTRL_SIZE; t += stride) ctrl[t] = 0.0;
+  if (Gw) {
+    for (long t = tid; t < total; t += stride) Gw[t] = 0.0;
+    grid_barrier(cnt, gen, nb);
+    // --- S0b: G assembly (edge-element stride loop, scatter-add) ----
+    const long ge = (long)g_ne * dh * R;
+    for (long i = tid; i < ge; i += stride) {
+      const int e = (int)(i / (dh * R));
+      const int t = (int)(i % (dh * R));
+      const int c = t / R, k = t % R;
+      const double* Ee = gE0 + (size_t)e * dh * dh;
+      const double* Xn = nbr + (size_t)g_nbr_slot[e] * dh * R;
+      double acc = 0.0;
+      #pragma unroll
+      for (int b = 0; b < dh; ++b)
+        acc = fma(Ee[c * dh + b], Xn[b * R + k], acc);
+      atomicAdd(&Gw[((size_t)g_local_pose[e] * dh + c) * R + k],
+                -g_w[e] * acc);
+    }
+  }
+  grid_barrier(cnt, gen, nb);
+  PSTAMP(1)
+
+  // --- S1: gradient phase (pose threads) + init tail ----------------
+  {
+    double d_fx = 0.0, d_gn = 0.0, d_gx = 0.0;
+    if (tid < n) {
+      double acc[dh][R];
+      #pragma unroll
+      for (int c = 0; c < dh; ++c)
+        #pragma unroll
+        for (int k = 0; k < R; ++k) acc[c][k] = 0.0;
+      const int s0 = q_rp[tid], e0 = q_rp[tid + 1];
+      for (int p = s0; p < e0; ++p) {
+        const double* B = q_vals + (size_t)p * dh * dh;
+        const double* Vj = X + (size_t)q_ci[p] * dh * R;
         #pragma unroll
         for (int c = 0; c < dh; ++c)
           #pragma unroll
-          for (int k = 0; k < R; ++k) acc[c][k] = 0.0;
-        const int s0 = q_rp[tid], e0 = q_rp[tid + 1];
-        for (int p = s0; p < e0; ++p) {
-          const double* B = q_vals + (size_t)p * dh * dh;
-          const double* Vj = delta + (size_t)q_ci[p] * dh * R;
-          #pragma unroll
-          for (int c = 0; c < dh; ++c)
-            #pragma unroll
-            for (int cc = 0; cc < dh; ++cc) {
-              const double b = B[c * dh + cc];
-              #pragma unroll
-              for (int k = 0; k < R; ++k)
-                acc[c][k] = fma(b, Vj[cc * R + k], acc[c][k]);
-            }
-        }
-        const double* Xi = X + (size_t)tid * dh * R;
-        double S[D][D];
-        #pragma unroll
-        for (int a = 0; a < D; ++a)
-          #pragma unroll
-          for (int b = 0; b < D; ++b) {
-            double sv = 0.0;
+          for (int cc = 0; cc < dh; ++cc) {
+            const double b = B[c * dh + cc];
             #pragma unroll
             for (int k = 0; k < R; ++k)
-              sv = fma(Xi[a * R + k], acc[b][k], sv);
-            S[a][b] = sv;
-          }
-        #pragma unroll
-        for (int a = 0; a < D; ++a)
-          #pragma unroll
-          for (int b = a; b < D; ++b) {
-            const double sv = 0.5 * (S[a][b] + S[b][a]);
-            S[a][b] = sv;
-            S[b][a] = sv;
-          }
-        #pragma unroll
-        for (int a = 0; a < D; ++a)
-          #pragma unroll
-          for (int k = 0; k < R; ++k) {
-            double v = acc[a][k];
-            #pragma unroll
-            for (int b = 0; b < D; ++b)
-              v = fma(-S[a][b], Xi[b * R + k], v);
-            acc[a][k] = v;
-          }
-        double* Oi = Hd + (size_t)tid * dh * R;
-        const double* Wi = delta + (size_t)tid * dh * R;
-        #pragma unroll
-        for (int c = 0; c < dh; ++c)
-          #pragma unroll
-          for (int k = 0; k < R; ++k) {
-            Oi[c * R + k] = acc[c][k];
-            d0 = fma(acc[c][k], Wi[c * R + k], d0);
+              acc[c][k] = fma(b, Vj[cc * R + k], acc[c][k]);
           }
       }
-      block_reduce_atomic(d0, ctrl + C_DOT0);
-      if (fanin_last_block(ctrl) && threadIdx.x == 0)
-        ctrl_tail_alpha(ctrl);
-    }
-    grid_barrier(cnt, gen, nb);
-
-    // --- stage U: eta/r update + snapshots, rr -> C_DOT1, tail_rr ---
-    {
-      const double coef = ctrl[C_COEF];
-      const int stop_pending = (int)ctrl[C_STOP_PENDING];
-      const int j = (int)ctrl[C_ITER];
-      double rr = 0.0;
-      if (tid < total) {
-        const double dl = delta[tid];
-        const double et = (j == 0) ? 0.0 : eta[tid];
-        eta_snap[(size_t)j * total + tid] = et;
-        delta_snap[(size_t)j * total + tid] = dl;
-        eta[tid] = fma(coef, dl, et);
-        if (!stop_pending) {
-          const double rn = fma(coef, Hd[tid], rvec[tid]);
-          rvec[tid] = rn;
-          rr = rn * rn;
+      const double* Xi = X + (size_t)tid * dh * R;
+      const double* Gi = G ? G + (size_t)tid * dh * R : nullptr;
+      #pragma unroll
+      for (int c = 0; c < dh; ++c)
+        #pragma unroll
+        for (int k = 0; k < R; ++k) {
+          double g = Gi ? Gi[c * R + k] : 0.0;
+          double v = acc[c][k] + g;
+          acc[c][k] = v;
+          d_fx = fma(v, Xi[c * R + k], d_fx);
+          d_gx = fma(g, Xi[c * R + k], d_gx);
         }
+      // tangent projection at X
+      double S[D][D];
+      #pragma unroll
+      for (int a = 0; a < D; ++a)
+        #pragma unroll
+        for (int b = 0; b < D; ++b) {
+          double sv = 0.0;
+          #pragma unroll
+          for (int k = 0; k < R; ++k)
+            sv = fma(Xi[a * R + k], acc[b][k], sv);
+          S[a][b] = sv;
+        }
+      #pragma unroll
+      for (int a = 0; a < D; ++a)
+        #pragma unroll
+        for (int b = a; b < D; ++b) {
+          const double sv = 0.5 * (S[a][b] + S[b][a]);
+          S[a][b] = sv;
+          S[b][a] = sv;
+        }
+      #pragma unroll
+      for (int a = 0; a < D; ++a)
+        #pragma unroll
+        for (int k = 0; k < R; ++k) {
+          double v = acc[a][k];
+          #pragma unroll
+          for (int b = 0; b < D; ++b)
+            v = fma(-S[a][b], Xi[b * R + k], v);
+          acc[a][k] = v;
+        }
+      double* Ri = rvec + (size_t)tid * dh * R;
+      #pragma unroll
+      for (int c = 0; c < dh; ++c)
+        #pragma unroll
+        for (int k = 0; k < R; ++k) {
+          Ri[c * R + k] = acc[c][k];
+          d_gn = fma(acc[c][k], acc[c][k], d_gn);
+        }
+    }
+    if (tid < total) z[tid] = 0.0;  // j-split precond accumulates
+    block_reduce_atomic(d_fx, ctrl + C_DOT0);
+    block_reduce_atomic(d_gn, ctrl + C_DOT1);
+    block_reduce_atomic(d_gx, ctrl + C_DOT2);
+    if (fanin_last_block(ctrl) && threadIdx.x == 0) {
+      // k_ctrl_init body
+      const double fX = 0.5 * (ctrl_load(ctrl + C_DOT0)
+                               + ctrl_load(ctrl + C_DOT2));
+      const double gn0sq = ctrl_load(ctrl + C_DOT1);
+      ctrl_store(ctrl + C_FX, fX);
+      ctrl_store(ctrl + C_GN0SQ, gn0sq);
+      ctrl_store(ctrl + C_RR, gn0sq);
+      const double norm_r0 = sqrt(gn0sq);
+      ctrl_store(ctrl + C_BOUND, norm_r0 * fmin(pow(norm_r0, theta), kappa));
+      ctrl_store(ctrl + C_RADIUS, Delta0);
+      ctrl_store(ctrl + C_EPE, 0.0);
+      ctrl_store(ctrl + C_EPD, 0.0);
+      ctrl_store(ctrl + C_ITER, 0.0);
+      ctrl_store(ctrl + C_J, 0.0);
+      ctrl_store(ctrl + C_USE_CURRENT, 0.0);
+      ctrl_store(ctrl + C_STOP_PENDING, 0.0);
+      ctrl_store(ctrl + C_BETA, 0.0);
+      ctrl_store(ctrl + C_STATUS, (norm_r0 < tol) ? (double)ST_NO_UPDATE
+                                                  : (double)ST_RUN);
+      ctrl_store(ctrl + C_DOT0, 0.0);
+      ctrl_store(ctrl + C_DOT1, 0.0);
+      ctrl_store(ctrl + C_DOT2, 0.0);
+      ctrl_store(ctrl + C_DOT3, 0.0);
+    }
+  }
+  grid_barrier(cnt, gen, nb);
+  }  // full: assembly + gradient phase
+  PSTAMP(2)
+
+  if (full && ctrl[C_STATUS] == (double)ST_RUN) {
+    // --- S2: z0 = P_X(Minv r0), <z0,r0> -> tail_z0; delta0 = -z0 ----
+    // j-split dense apply: the extra (jmul-1)*total threads of the
+    // widened grid each take a chunk of the j-range (the apply is
+    // latency-bound on the L2-cold pass over Minv, so memory-level
+    // parallelism is the lever); 4 accumulator chains per thread.
+    if (tid < (long)total * jmul) {
+      const long e = tid % total;
+      const int q = (int)(tid / total);
+      const int i = (int)(e / R), k = (int)(e % R);
+      const int jchunk = (N + jmul - 1) / jmul;
+      const int jlo = q * jchunk, jhi = min(N, jlo + jchunk);
+      double a0 = 0.0, a1 = 0.0, a2 = 0.0, a3 = 0.0;
+      int jj = jlo;
+      for (; jj + 4 <= jhi; jj += 4) {
+        a0 = fma((double)Minv[(size_t)(jj + 0) * N + i],
+                 rvec[(size_t)(jj + 0) * R + k], a0);
+        a1 = fma((double)Minv[(size_t)(jj + 1) * N + i],
+                 rvec[(size_t)(jj + 1) * R + k], a1);
+        a2 = fma((double)Minv[(size_t)(jj + 2) * N + i],
+                 rvec[(size_t)(jj + 2) * R + k], a2);
+        a3 = fma((double)Minv[(size_t)(jj + 3) * N + i],
+                 rvec[(size_t)(jj + 3) * R + k], a3);
       }
-      if (!stop_pending) block_reduce_atomic(rr, ctrl + C_DOT1);
-      if (fanin_last_block(ctrl) && threadIdx.x == 0)
-        ctrl_tail_rr(ctrl);
+      for (; jj < jhi; ++jj)
+        a0 = fma((double)Minv[(size_t)jj * N + i],
+                 rvec[(size_t)jj * R + k], a0);
+      const double part = (a0 + a1) + (a2 + a3);
+      if (jmul == 1) z[e] = part;
+      else atomicAdd(&z[e], part);
     }
     grid_barrier(cnt, gen, nb);
-    if (ctrl[C_STATUS] != (double)ST_RUN) break;
-
-    // --- stage P: z = Minv r (dense fp32 apply, one element/thread) -
-    if (tid < total) {
-      const int i = tid / R, k = tid % R;
-      double acc = 0.0;
-      for (int jj = 0; jj < N; ++jj)
-        acc = fma((double)Minv[(size_t)jj * N + i], rvec[(size_t)jj * R + k],
-                  acc);
-      z[tid] = acc;
-    }
-    grid_barrier(cnt, gen, nb);
-
-    // --- stage J: project z at X, <z, r> -> C_DOT0, tail_beta -------
     {
       double d0 = 0.0;
       if (tid < n) {
@@ -1048,8 +1155,7 @@ __global__ void k_tcg_persist(const int* __restrict__ q_rp,
           for (int b = 0; b < D; ++b) {
             double sv = 0.0;
             #pragma unroll
-            for (int k = 0; k < R; ++k)
-              sv = fma(Xi[a * R + k], Vt[b][k], sv);
+            for (int k = 0; k < R; ++k) sv = fma(Xi[a * R + k], Vt[b][k], sv);
             S[a][b] = sv;
           }
         #pragma unroll
@@ -1080,14 +1186,328 @@ __global__ void k_tcg_persist(const int* __restrict__ q_rp,
       }
       block_reduce_atomic(d0, ctrl + C_DOT0);
       if (fanin_last_block(ctrl) && threadIdx.x == 0)
-        ctrl_tail_beta(ctrl);
+        ctrl_tail_z0(ctrl);
     }
     grid_barrier(cnt, gen, nb);
+    if (tid < total) delta[tid] = -z[tid];
+  }  // full: z0 phase
+  PSTAMP(3)
 
-    // --- stage D: delta = beta*delta - z ----------------------------
-    if (tid < total)
-      delta[tid] = fma(ctrl[C_BETA], delta[tid], -z[tid]);
+  {
+    // --- S3: the tCG loop -------------------------------------------
+    for (int it = 0; it < max_inner; ++it) {
+      grid_barrier(cnt, gen, nb);  // prior delta writes visible
+      if (ctrl[C_STATUS] != (double)ST_RUN) break;
+
+      // H: Hd = P_X(Q delta), d_Qd -> C_DOT0, tail_alpha
+      {
+        double d0 = 0.0;
+        if (tid < n) {
+          double acc[dh][R];
+          #pragma unroll
+          for (int c = 0; c < dh; ++c)
+            #pragma unroll
+            for (int k = 0; k < R; ++k) acc[c][k] = 0.0;
+          const int s0 = q_rp[tid], e0 = q_rp[tid + 1];
+          for (int p = s0; p < e0; ++p) {
+            const double* B = q_vals + (size_t)p * dh * dh;
+            const double* Vj = delta + (size_t)q_ci[p] * dh * R;
+            #pragma unroll
+            for (int c = 0; c < dh; ++c)
+              #pragma unroll
+              for (int cc = 0; cc < dh; ++cc) {
+                const double b = B[c * dh + cc];
+                #pragma unroll
+                for (int k = 0; k < R; ++k)
+                  acc[c][k] = fma(b, Vj[cc * R + k], acc[c][k]);
+              }
+          }
+          const double* Xi = X + (size_t)tid * dh * R;
+          double S[D][D];
+          #pragma unroll
+          for (int a = 0; a < D; ++a)
+            #pragma unroll
+            for (int b = 0; b < D; ++b) {
+              double sv = 0.0;
+              #pragma unroll
+              for (int k = 0; k < R; ++k)
+                sv = fma(Xi[a * R + k], acc[b][k], sv);
+              S[a][b] = sv;
+            }
+          #pragma unroll
+          for (int a = 0; a < D; ++a)
+            #pragma unroll
+            for (int b = a; b < D; ++b) {
+              const double sv = 0.5 * (S[a][b] + S[b][a]);
+              S[a][b] = sv;
+              S[b][a] = sv;
+            }
+          #pragma unroll
+          for (int a = 0; a < D; ++a)
+            #pragma unroll
+            for (int k = 0; k < R; ++k) {
+              double v = acc[a][k];
+              #pragma unroll
+              for (int b = 0; b < D; ++b)
+                v = fma(-S[a][b], Xi[b * R + k], v);
+              acc[a][k] = v;
+            }
+          double* Oi = Hd + (size_t)tid * dh * R;
+          const double* Wi = delta + (size_t)tid * dh * R;
+          #pragma unroll
+          for (int c = 0; c < dh; ++c)
+            #pragma unroll
+            for (int k = 0; k < R; ++k) {
+              Oi[c * R + k] = acc[c][k];
+              d0 = fma(acc[c][k], Wi[c * R + k], d0);
+            }
+        }
+        block_reduce_atomic(d0, ctrl + C_DOT0);
+        if (fanin_last_block(ctrl) && threadIdx.x == 0)
+          ctrl_tail_alpha(ctrl);
+      }
+      grid_barrier(cnt, gen, nb);
+
+      // U: eta/r update + snapshots, rr -> C_DOT1, tail_rr
+      {
+        const double coef = ctrl[C_COEF];
+        const int stop_pending = (int)ctrl[C_STOP_PENDING];
+        const int j = (int)ctrl[C_ITER];
+        double rr = 0.0;
+        if (tid < total) z[tid] = 0.0;  // next j-split apply accumulates
+        if (tid < total) {
+          const double dl = delta[tid];
+          const double et = (j == 0) ? 0.0 : eta[tid];
+          eta_snap[(size_t)j * total + tid] = et;
+          delta_snap[(size_t)j * total + tid] = dl;
+          eta[tid] = fma(coef, dl, et);
+          if (!stop_pending) {
+            const double rn = fma(coef, Hd[tid], rvec[tid]);
+            rvec[tid] = rn;
+            rr = rn * rn;
+          }
+        }
+        if (!stop_pending) block_reduce_atomic(rr, ctrl + C_DOT1);
+        if (fanin_last_block(ctrl) && threadIdx.x == 0)
+          ctrl_tail_rr(ctrl);
+      }
+      grid_barrier(cnt, gen, nb);
+      if (ctrl[C_STATUS] != (double)ST_RUN) break;
+
+      // P: z = Minv r (j-split dense fp32 apply; see z0 stage)
+      if (tid < (long)total * jmul) {
+        const long e = tid % total;
+        const int q = (int)(tid / total);
+        const int i = (int)(e / R), k = (int)(e % R);
+        const int jchunk = (N + jmul - 1) / jmul;
+        const int jlo = q * jchunk, jhi = min(N, jlo + jchunk);
+        double a0 = 0.0, a1 = 0.0, a2 = 0.0, a3 = 0.0;
+        int jj = jlo;
+        for (; jj + 4 <= jhi; jj += 4) {
+          a0 = fma((double)Minv[(size_t)(jj + 0) * N + i],
+                   rvec[(size_t)(jj + 0) * R + k], a0);
+          a1 = fma((double)Minv[(size_t)(jj + 1) * N + i],
+                   rvec[(size_t)(jj + 1) * R + k], a1);
+          a2 = fma((double)Minv[(size_t)(jj + 2) * N + i],
+                   rvec[(size_t)(jj + 2) * R + k], a2);
+          a3 = fma((double)Minv[(size_t)(jj + 3) * N + i],
+                   rvec[(size_t)(jj + 3) * R + k], a3);
+        }
+        for (; jj < jhi; ++jj)
+          a0 = fma((double)Minv[(size_t)jj * N + i],
+                   rvec[(size_t)jj * R + k], a0);
+        const double part = (a0 + a1) + (a2 + a3);
+        if (jmul == 1) z[e] = part;
+        else atomicAdd(&z[e], part);
+      }
+      grid_barrier(cnt, gen, nb);
+
+      // J: project z at X, <z, r> -> C_DOT0, tail_beta
+      {
+        double d0 = 0.0;
+        if (tid < n) {
+          const double* Xi = X + (size_t)tid * dh * R;
+          double* Zi = z + (size_t)tid * dh * R;
+          const double* Ri = rvec + (size_t)tid * dh * R;
+          double Vt[dh][R];
+          #pragma unroll
+          for (int c = 0; c < dh; ++c)
+            #pragma unroll
+            for (int k = 0; k < R; ++k) Vt[c][k] = Zi[c * R + k];
+          double S[D][D];
+          #pragma unroll
+          for (int a = 0; a < D; ++a)
+            #pragma unroll
+            for (int b = 0; b < D; ++b) {
+              double sv = 0.0;
+              #pragma unroll
+              for (int k = 0; k < R; ++k)
+                sv = fma(Xi[a * R + k], Vt[b][k], sv);
+              S[a][b] = sv;
+            }
+          #pragma unroll
+          for (int a = 0; a < D; ++a)
+            #pragma unroll
+            for (int b = a; b < D; ++b) {
+              const double sv = 0.5 * (S[a][b] + S[b][a]);
+              S[a][b] = sv;
+              S[b][a] = sv;
+            }
+          #pragma unroll
+          for (int a = 0; a < D; ++a)
+            #pragma unroll
+            for (int k = 0; k < R; ++k) {
+              double v = Vt[a][k];
+              #pragma unroll
+              for (int b = 0; b < D; ++b)
+                v = fma(-S[a][b], Xi[b * R + k], v);
+              Vt[a][k] = v;
+            }
+          #pragma unroll
+          for (int c = 0; c < dh; ++c)
+            #pragma unroll
+            for (int k = 0; k < R; ++k) {
+              Zi[c * R + k] = Vt[c][k];
+              d0 = fma(Vt[c][k], Ri[c * R + k], d0);
+            }
+        }
+        block_reduce_atomic(d0, ctrl + C_DOT0);
+        if (fanin_last_block(ctrl) && threadIdx.x == 0)
+          ctrl_tail_beta(ctrl);
+      }
+      grid_barrier(cnt, gen, nb);
+
+      // D: delta = beta*delta - z
+      if (tid < total)
+        delta[tid] = fma(ctrl[C_BETA], delta[tid], -z[tid]);
+    }
+    grid_barrier(cnt, gen, nb);
+    PSTAMP(4)
+    if (full && blockIdx.x == 0 && threadIdx.x == 0) {
+      // k_ctrl_tcg_end body
+      if (ctrl[C_STATUS] == (double)ST_RUN) {
+        ctrl_store(ctrl + C_STATUS, (double)ST_TCG_STOP);
+        ctrl_store(ctrl + C_J, ctrl[C_ITER]);
+        ctrl_store(ctrl + C_HLEN, ctrl[C_ITER]);
+        ctrl_store(ctrl + C_USE_CURRENT, 1.0);
+      }
+      ctrl_candidate_logic(ctrl);
+    }
+    grid_barrier(cnt, gen, nb);
   }
+
+  // --- S4: first candidate: form step, retract, evaluate, accept ----
+  PSTAMP(5)
+  if (full && ctrl[C_STATUS] == (double)ST_TCG_STOP) {
+    if (tid < total) {
+      if (ctrl[C_USE_CURRENT] != 0.0) {
+        step[tid] = eta[tid];
+      } else {
+        const int j = (int)ctrl[C_JSTAR];
+        const double tau = ctrl[C_TAUSTAR];
+        step[tid] = fma(tau, delta_snap[(size_t)j * total + tid],
+                        eta_snap[(size_t)j * total + tid]);
+      }
+    }
+    grid_barrier(cnt, gen, nb);
+    // polar retraction Xprop = polar(X + step)
+    if (tid < n) {
+      double Mt[dh][R];
+      const double* Ai = X + (size_t)tid * dh * R;
+      const double* Bi = step + (size_t)tid * dh * R;
+      #pragma unroll
+      for (int c = 0; c < dh; ++c)
+        #pragma unroll
+        for (int k = 0; k < R; ++k) Mt[c][k] = Ai[c * R + k] + Bi[c * R + k];
+      double S[D][D];
+      #pragma unroll
+      for (int a = 0; a < D; ++a)
+        #pragma unroll
+        for (int b = 0; b < D; ++b) {
+          double s = 0.0;
+          #pragma unroll
+          for (int k = 0; k < R; ++k) s = fma(Mt[a][k], Mt[b][k], s);
+          S[a][b] = s;
+        }
+      double Gi2[D][D];
+      spd_inv_sqrt<D>(S, Gi2);
+      double* Oi = Xprop + (size_t)tid * dh * R;
+      #pragma unroll
+      for (int a = 0; a < D; ++a)
+        #pragma unroll
+        for (int k = 0; k < R; ++k) {
+          double s = 0.0;
+          #pragma unroll
+          for (int b = 0; b < D; ++b) s = fma(Gi2[a][b], Mt[b][k], s);
+          Oi[a * R + k] = s;
+        }
+      #pragma unroll
+      for (int k = 0; k < R; ++k) Oi[D * R + k] = Mt[D][k];
+    }
+    grid_barrier(cnt, gen, nb);
+    // f(Xprop) dots: <Q Xprop, Xprop> -> C_DOT0, <G, Xprop> -> C_DOT2
+    {
+      double d0 = 0.0, d1 = 0.0;
+      if (tid < n) {
+        double acc[dh][R];
+        #pragma unroll
+        for (int c = 0; c < dh; ++c)
+          #pragma unroll
+          for (int k = 0; k < R; ++k) acc[c][k] = 0.0;
+        const int s0 = q_rp[tid], e0 = q_rp[tid + 1];
+        for (int p = s0; p < e0; ++p) {
+          const double* B = q_vals + (size_t)p * dh * dh;
+          const double* Vj = Xprop + (size_t)q_ci[p] * dh * R;
+          #pragma unroll
+          for (int c = 0; c < dh; ++c)
+            #pragma unroll
+            for (int cc = 0; cc < dh; ++cc) {
+              const double b = B[c * dh + cc];
+              #pragma unroll
+              for (int k = 0; k < R; ++k)
+                acc[c][k] = fma(b, Vj[cc * R + k], acc[c][k]);
+            }
+        }
+        const double* Vi = Xprop + (size_t)tid * dh * R;
+        const double* Gi = G ? G + (size_t)tid * dh * R : nullptr;
+        #pragma unroll
+        for (int c = 0; c < dh; ++c)
+          #pragma unroll
+          for (int k = 0; k < R; ++k) {
+            const double g = Gi ? Gi[c * R + k] : 0.0;
+            d0 = fma(acc[c][k], Vi[c * R + k], d0);
+            d1 = fma(g, Vi[c * R + k], d1);
+          }
+      }
+      block_reduce_atomic(d0, ctrl + C_DOT0);
+      block_reduce_atomic(d1, ctrl + C_DOT2);
+      if (fanin_last_block(ctrl) && threadIdx.x == 0) {
+        // k_ctrl_accept body
+        if (ctrl[C_STATUS] == (double)ST_TCG_STOP) {
+          const double fprop = 0.5 * ctrl_load(ctrl + C_DOT0)
+                               + ctrl_load(ctrl + C_DOT2);
+          ctrl_store(ctrl + C_DOT0, 0.0);
+          ctrl_store(ctrl + C_DOT2, 0.0);
+          ctrl_store(ctrl + C_FPROP, fprop);
+          const double fX = ctrl[C_FX];
+          const double dm = ctrl[C_DM];
+          const double rho = (fX - fprop) / fmax(dm, 1e-300);
+          ctrl_store(ctrl + C_RHO, rho);
+          if (rho > accept_rho && fprop <= fX)
+            ctrl_store(ctrl + C_STATUS, (double)ST_ACCEPTED);
+        }
+      }
+    }
+  }
+  grid_barrier(cnt, gen, nb);
+  PSTAMP(6)
+
+  // --- S5: publish the control block to the host --------------------
+  if (full && blockIdx.x == 0)
+    for (int t = threadIdx.x; t < CTRL_SIZE; t += blockDim.x)
+      ctrl_host[t] = ctrl_load(ctrl + t);
+  PSTAMP(7)
+#undef PSTAMP
 }
 
 // ---------------------------------------------------------------------
@@ -1882,7 +2302,8 @@ void* dpo_ctx_create(int n, int d, int r, int max_inner) {
   // per-solve zeroing passes are dropped from the captured body
   DPO_CHECK(hipMemset(c->delta, 0, vb));
   DPO_CHECK(hipMemset(c->eta, 0, vb));
-  DPO_CHECK(hipHostMalloc(&c->ctrl_host, CTRL_SIZE * sizeof(double)));
+  // +16 slots: persistent-kernel stage timestamps (DPO_DBG_PERSIST=1)
+  DPO_CHECK(hipHostMalloc(&c->ctrl_host, (CTRL_SIZE + 16) * sizeof(double)));
   DPO_CHECK(hipHostGetDevicePointer((void**)&c->ctrl_host_dev,
                                     c->ctrl_host, 0));
   DPO_CHECK(hipStreamCreateWithFlags(&c->cap_stream,
@@ -1927,15 +2348,45 @@ void dpo_ctx_set_problem(void* h, const int* rp, const int* ci,
   c->Gt = Gt; c->Minv = Minv; c->Ljac = Ljac;
 }
 
-static void launch_tcg_persist(DpoCtx* c, const double* X, hipStream_t s) {
+static void launch_solve_persist(DpoCtx* c, const double* X,
+                                 const double* nbr, double tol,
+                                 double Delta0, double accept_rho,
+                                 hipStream_t s, int full) {
   const int gvec = (int)((c->total + 255) / 256);
+  // replicate ctx_assemble_g's pointer logic (assembly itself runs
+  // inside the kernel); c->Gt must be correct for the host-driven
+  // shrink-loop replay in solve_postsync
+  const double* G;
+  double* Gw = nullptr;
+  if (nbr) {
+    if (c->g_ne) {
+      c->Gt = c->G_buf;
+      G = c->G_buf;
+      Gw = c->G_buf;
+    } else {
+      c->Gt = nullptr;
+      G = nullptr;
+    }
+  } else {
+    G = c->Gt;
+  }
+  // NOTE: widening the grid for j-split preconditioner stages was
+  // measured SLOWER under multi-agent concurrency (4 agents x 200
+  // workgroups oversubscribes the 256 CUs and the idle blocks' barrier
+  // spins starve real work) — keep the grid at one block per 256
+  // elements and rely on the 4-chain unroll for MLP.
+  const int jmul = 1;
 #define CASE_TP(D, R) \
   if (c->d == D && c->r == R) { \
-    hipLaunchKernelGGL((k_tcg_persist<D, R>), dim3(gvec), dim3(256), 0, s, \
-                       c->q_rp, c->q_ci, c->q_vals, X, c->Minv, c->eta, \
-                       c->rvec, c->delta, c->z, c->Hd, c->eta_snap, \
-                       c->delta_snap, c->ctrl, c->gbar, c->n, c->N, \
-                       c->total, c->max_inner); \
+    hipLaunchKernelGGL((k_solve_persist<D, R>), dim3(gvec * jmul), \
+                       dim3(256), 0, \
+                       s, c->q_rp, c->q_ci, c->q_vals, X, G, Gw, c->g_E0, \
+                       c->g_local_pose, c->g_nbr_slot, nbr, c->g_w, \
+                       c->g_ne, c->Minv, c->eta, c->rvec, c->delta, c->z, \
+                       c->Hd, c->eta_snap, c->delta_snap, c->step, \
+                       c->Xprop, c->ctrl, c->ctrl_host_dev, c->gbar, \
+                       c->n, c->N, c->total, c->max_inner, tol, Delta0, \
+                       1.0, 0.1, accept_rho, jmul, full); \
     return; \
   }
   DPO_FOREACH_DR(CASE_TP)
@@ -1962,7 +2413,23 @@ static void enqueue_solve_body(DpoCtx* c, double* X, const double* nbr,
   // block_reduce_atomic plus the hipGraph ordering bugs (both fixed);
   // DPO_NO_CF=1 restores the dedicated control kernels.
   static const bool no_cf = dpo_env_flag("DPO_NO_CF");
+  static const bool no_persist = dpo_env_flag("DPO_NO_PERSIST");
   fence_wait(c, F_SOLVE_IN, s);  // first node: block until inputs ready
+  // Persistent-kernel scope: by default the tCG LOOP runs as one
+  // kernel (k_solve_persist with full=0) between the grad/z0 and
+  // candidate launch sequences — measured fastest under multi-agent
+  // concurrency. DPO_PERSIST_FULL=1 fuses the ENTIRE pre-sync solve
+  // into the kernel (fewer launches but the in-kernel z0 dense-
+  // preconditioner apply is latency-bound on the L2-cold Minv pass,
+  // which the standalone j-split kernel hides better).
+  static const bool persist_full = dpo_env_flag("DPO_PERSIST_FULL");
+  const bool persist_ok = !no_persist && !no_cf && c->Minv != nullptr
+      && (total + 255) / 256 <= 256;
+  if (persist_ok && persist_full) {
+    launch_solve_persist(c, X, nbr, tol, Delta0, accept_rho, s, 1);
+    fence_signal(c, F_SOLVE_OUT, s);
+    return;
+  }
   if (nbr) ctx_assemble_g(c, nbr, s);
   dzero(c->ctrl, CTRL_SIZE, s);
 
@@ -1992,17 +2459,10 @@ static void enqueue_solve_body(DpoCtx* c, double* X, const double* nbr,
   hipLaunchKernelGGL(k_tcg_delta, dim3(gvec), dim3(256), 0, s,
                      c->delta, c->z, c->ctrl, total);
 
-  // Persistent whole-loop kernel when the grid is small enough for
-  // guaranteed co-residency (<= 256 workgroups on 256 CUs) and the
-  // dense preconditioner is bound; DPO_NO_PERSIST=1 falls back to the
-  // per-stage launch sequence below.
-  static const bool no_persist = dpo_env_flag("DPO_NO_PERSIST");
-  if (!no_persist && !no_cf && c->Minv != nullptr && gvec <= 256) {
-    launch_tcg_persist(c, X, s);
-    hipLaunchKernelGGL(k_ctrl_tcg_end, dim3(1), dim3(64), 0, s, c->ctrl);
-    goto candidate;
-  }
-
+  if (persist_ok) {
+    // the whole tCG loop as one kernel (grid barriers between stages)
+    launch_solve_persist(c, X, nbr, tol, Delta0, accept_rho, s, 0);
+  } else
   for (int j = 0; j < c->max_inner; ++j) {
     if (no_cf) {
       launch_hess_fused<0>(c->q_rp, c->q_ci, c->q_vals, c->delta, X,
@@ -2035,7 +2495,6 @@ static void enqueue_solve_body(DpoCtx* c, double* X, const double* nbr,
                        c->delta, c->z, c->ctrl, total);
   }
   hipLaunchKernelGGL(k_ctrl_tcg_end, dim3(1), dim3(64), 0, s, c->ctrl);
-candidate:
   // first candidate attempt is part of the fixed sequence
   hipLaunchKernelGGL(k_ctrl_candidate, dim3(1), dim3(64), 0, s, c->ctrl);
   hipLaunchKernelGGL(k_form_step, dim3(gvec), dim3(256), 0, s,
@@ -2239,6 +2698,14 @@ int dpo_round_solve_finish(void* h, int max_shrink, double* stats_out,
                            void* join_stream) {
   DpoCtx* c = (DpoCtx*)h;
   DPO_CHECK(hipStreamSynchronize(c->exec_stream));
+  if (dpo_env_flag("DPO_DBG_PERSIST")) {
+    const double* ts = c->ctrl_host + CTRL_SIZE;
+    fprintf(stderr, "[persist us] zeroG %.1f grad %.1f z0 %.1f loop %.1f"
+            " end %.1f cand %.1f pub %.1f\n",
+            (ts[1]-ts[0])/100.0, (ts[2]-ts[1])/100.0, (ts[3]-ts[2])/100.0,
+            (ts[4]-ts[3])/100.0, (ts[5]-ts[4])/100.0, (ts[6]-ts[5])/100.0,
+            (ts[7]-ts[6])/100.0);
+  }
   int st = solve_postsync(c, c->pend_X, c->pend_rho, max_shrink, 0,
                           stats_out, c->exec_stream);
   DPO_CHECK(hipEventRecord(c->done_event, c->exec_stream));
