@@ -2525,10 +2525,10 @@ static bool solve_presync(DpoCtx* c, double* X, const double* nbr,
                           hipStream_t s) {
   static const bool no_graph = dpo_env_flag("DPO_NO_SOLVE_GRAPH");
   if (no_graph) {
+    // fully async eager enqueue; callers sync (finish / impl)
     enqueue_solve_body(c, X, nbr, tol, Delta0, accept_rho, s);
     fence_wait(c, F_SOLVE_OUT, s);
-    DPO_CHECK(hipStreamSynchronize(s));
-    return false;
+    return true;
   }
   const void* key[4] = {X, nbr, (const void*)(intptr_t)(tol * 1e9),
                         (const void*)(intptr_t)Delta0};
@@ -2670,7 +2670,8 @@ int dpo_rbcd_solve(void* h, double* X, double tol, double Delta0,
 void dpo_round_eval(void* h, const double* X, const double* nbr,
                     double* out_dev, void* stream);  // fwd decl
 static void round_eval_impl(DpoCtx* c, const double* X, const double* nbr,
-                            double* out_dev, hipStream_t s);  // fwd decl
+                            double* out_dev, hipStream_t s,
+                            bool wait_out = true);  // fwd decl
 
 // --- async (multi-stream) round entry points ------------------------
 // Launch the solve's pre-sync sequence on the ctx's private execution
@@ -2683,10 +2684,10 @@ void dpo_round_solve_async(void* h, double* X, const double* nbr,
   hipStream_t js = (hipStream_t)join_stream;
   // the IN fence is signalled on the PRODUCER (torch) stream: it
   // carries the completion of the boundary-pose scatter into the solve
-  // sequence as a data dependency (see fence comment above)
+  // sequence as a data dependency (see fence comment above). No event
+  // ceremony — the fence IS the ordering mechanism (launch-order
+  // hipStreamWaitEvent is exactly what graph launches fail to honor).
   fence_signal(c, F_SOLVE_IN, js);
-  DPO_CHECK(hipEventRecord(c->start_event, js));
-  DPO_CHECK(hipStreamWaitEvent(c->exec_stream, c->start_event, 0));
   solve_presync(c, X, nbr, tol, Delta0, accept_rho, c->exec_stream);
   c->pend_X = X;
   c->pend_tol = tol;
@@ -2719,9 +2720,7 @@ void dpo_round_eval_async(void* h, const double* X, const double* nbr,
   hipStream_t js = (hipStream_t)join_stream;
   // IN fence on the producer stream: carries the boundary-pose scatter
   fence_signal(c, F_EVAL_IN, js);
-  DPO_CHECK(hipEventRecord(c->start_event, js));
-  DPO_CHECK(hipStreamWaitEvent(c->exec_stream, c->start_event, 0));
-  round_eval_impl(c, X, nbr, out_dev, c->exec_stream);
+  round_eval_impl(c, X, nbr, out_dev, c->exec_stream, true);
   DPO_CHECK(hipEventRecord(c->done_event, c->exec_stream));
 }
 
@@ -2800,13 +2799,16 @@ static void enqueue_eval_body(DpoCtx* c, const double* X, const double* nbr,
   fence_signal(c, F_EVAL_OUT, s);
 }
 
-// IN fence must already be signalled on the producer stream.
+// IN fence must already be signalled on the producer stream. With
+// wait_out=false the caller consumes the OUT fence itself (group
+// fan-out path: one gather kernel waits on every agent's OUT fence).
 static void round_eval_impl(DpoCtx* c, const double* X, const double* nbr,
-                            double* out_dev, hipStream_t s) {
+                            double* out_dev, hipStream_t s,
+                            bool wait_out) {
   static const bool no_graph = dpo_env_flag("DPO_NO_EVAL_GRAPH");
   if (no_graph) {
     enqueue_eval_body(c, X, nbr, out_dev, s);
-    fence_wait(c, F_EVAL_OUT, s);
+    if (wait_out) fence_wait(c, F_EVAL_OUT, s);
     return;
   }
   const void* key[4] = {X, nbr, out_dev, nullptr};
@@ -2841,14 +2843,14 @@ static void round_eval_impl(DpoCtx* c, const double* X, const double* nbr,
     if (rc != hipSuccess || !c->eval_graph) {
       c->eval_graph = nullptr;
       enqueue_eval_body(c, X, nbr, out_dev, s);
-      fence_wait(c, F_EVAL_OUT, s);
+      if (wait_out) fence_wait(c, F_EVAL_OUT, s);
       return;
     }
     memcpy(c->eval_key, key, sizeof(key));
   }
   c->eval_replays++;
   DPO_CHECK(hipGraphLaunch(c->eval_graph, s));
-  fence_wait(c, F_EVAL_OUT, s);
+  if (wait_out) fence_wait(c, F_EVAL_OUT, s);
 }
 
 void dpo_round_eval(void* h, const double* X, const double* nbr,
@@ -2875,6 +2877,40 @@ __global__ void k_gather3(double* __restrict__ out, long stride,
   if (i < n && j < 3) out[(long)rows[i] * stride + j] = srcs[i][j];
 }
 
+// Signal every agent's eval IN fence with one launch (lane per agent).
+__global__ void k_fence_signal_many(unsigned int* const* __restrict__ f,
+                                    int n) {
+  int i = threadIdx.x;
+  if (i < n)
+    __hip_atomic_store(f[i], 1u, __ATOMIC_RELEASE,
+                       __HIP_MEMORY_SCOPE_AGENT);
+}
+
+// Wait on & consume every agent's eval OUT fence, then gather each
+// agent's 3 eval scalars (agent-scope loads: the producers ran on
+// other XCDs) — replaces n (event-record + event-wait + copy) chains.
+__global__ void k_gather3_fenced(double* __restrict__ out, long stride,
+                                 double* const* __restrict__ srcs,
+                                 const int* __restrict__ rows,
+                                 unsigned int* const* __restrict__ f,
+                                 int n) {
+  int i = threadIdx.x;
+  if (i < n) {
+    long spins = 0;
+    while (__hip_atomic_load(f[i], __ATOMIC_ACQUIRE,
+                             __HIP_MEMORY_SCOPE_AGENT) == 0u) {
+      __builtin_amdgcn_s_sleep(32);
+      if (++spins > 30000000L) __builtin_trap();
+    }
+    __hip_atomic_store(f[i], 0u, __ATOMIC_RELAXED,
+                       __HIP_MEMORY_SCOPE_AGENT);
+    #pragma unroll
+    for (int j = 0; j < 3; ++j)
+      out[(long)rows[i] * stride + j] = __hip_atomic_load(
+          srcs[i] + j, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+  }
+}
+
 struct DpoGroup {
   int n;
   DpoCtx** cs;
@@ -2882,6 +2918,8 @@ struct DpoGroup {
   const double** nbrs;
   double** d_srcs;      // device array: each ctx's eval3
   int* d_rows;          // device array: output row per agent
+  unsigned int** d_fin;   // device array: each ctx's F_EVAL_IN fence
+  unsigned int** d_fout;  // device array: each ctx's F_EVAL_OUT fence
 };
 
 void dpo_group_destroy(void* g) {
@@ -2889,6 +2927,8 @@ void dpo_group_destroy(void* g) {
   if (!gr) return;
   hipFree(gr->d_srcs);
   hipFree(gr->d_rows);
+  hipFree(gr->d_fin);
+  hipFree(gr->d_fout);
   free(gr->cs); free(gr->Xs); free(gr->nbrs);
   delete gr;
 }
@@ -2902,17 +2942,27 @@ void* dpo_group_create(void** handles, int n, double** Xs,
   gr->Xs = (double**)malloc(n * sizeof(void*));
   gr->nbrs = (const double**)malloc(n * sizeof(void*));
   double* srcs_h[256];
+  unsigned int* fin_h[256];
+  unsigned int* fout_h[256];
   for (int i = 0; i < n; ++i) {
     gr->cs[i] = (DpoCtx*)handles[i];
     gr->Xs[i] = Xs[i];
     gr->nbrs[i] = nbrs[i];
     srcs_h[i] = gr->cs[i]->eval3;
+    fin_h[i] = gr->cs[i]->fences + F_EVAL_IN;
+    fout_h[i] = gr->cs[i]->fences + F_EVAL_OUT;
   }
   DPO_CHECK(hipMalloc(&gr->d_srcs, n * sizeof(double*)));
   DPO_CHECK(hipMemcpy(gr->d_srcs, srcs_h, n * sizeof(double*),
                       hipMemcpyHostToDevice));
   DPO_CHECK(hipMalloc(&gr->d_rows, n * sizeof(int)));
   DPO_CHECK(hipMemcpy(gr->d_rows, rows, n * sizeof(int),
+                      hipMemcpyHostToDevice));
+  DPO_CHECK(hipMalloc(&gr->d_fin, n * sizeof(void*)));
+  DPO_CHECK(hipMemcpy(gr->d_fin, fin_h, n * sizeof(void*),
+                      hipMemcpyHostToDevice));
+  DPO_CHECK(hipMalloc(&gr->d_fout, n * sizeof(void*)));
+  DPO_CHECK(hipMemcpy(gr->d_fout, fout_h, n * sizeof(void*),
                       hipMemcpyHostToDevice));
   return gr;
 }
@@ -2945,13 +2995,18 @@ void dpo_group_eval(void* g, double* out_dev, long row_stride,
                     void* join_stream) {
   DpoGroup* gr = (DpoGroup*)g;
   hipStream_t js = (hipStream_t)join_stream;
+  // one kernel signals every agent's IN fence (carries the boundary
+  // scatter ordering); the per-agent eval graphs are then launched
+  // bare on their exec streams — no event ceremony — and one gather
+  // kernel on the join stream waits on & consumes every OUT fence.
+  hipLaunchKernelGGL(k_fence_signal_many, dim3(1), dim3(256), 0, js,
+                     (unsigned int* const*)gr->d_fin, gr->n);
   for (int i = 0; i < gr->n; ++i)
-    dpo_round_eval_async(gr->cs[i], gr->Xs[i], gr->nbrs[i],
-                         gr->cs[i]->eval3, join_stream);
-  for (int i = 0; i < gr->n; ++i)
-    DPO_CHECK(hipStreamWaitEvent(js, gr->cs[i]->done_event, 0));
-  hipLaunchKernelGGL(k_gather3, dim3(gr->n), dim3(64), 0, js,
-                     out_dev, row_stride, gr->d_srcs, gr->d_rows, gr->n);
+    round_eval_impl(gr->cs[i], gr->Xs[i], gr->nbrs[i], gr->cs[i]->eval3,
+                    gr->cs[i]->exec_stream, /*wait_out=*/false);
+  hipLaunchKernelGGL(k_gather3_fenced, dim3(1), dim3(256), 0, js,
+                     out_dev, row_stride, gr->d_srcs, gr->d_rows,
+                     (unsigned int* const*)gr->d_fout, gr->n);
 }
 
 }  // extern "C"
