@@ -2423,8 +2423,14 @@ static void enqueue_solve_body(DpoCtx* c, double* X, const double* nbr,
   // preconditioner apply is latency-bound on the L2-cold Minv pass,
   // which the standalone j-split kernel hides better).
   static const bool persist_full = dpo_env_flag("DPO_PERSIST_FULL");
+  // N cap: the in-kernel dense preconditioner apply (one element per
+  // thread, 4 load chains) only beats the standalone j-split kernel
+  // while Minv stays cache-resident; for larger problems the
+  // per-stage launch path's j-split apply sustains far higher
+  // bandwidth (measured: city10000/cubicle agents at N ~ 4600-6000
+  // were 4-8x slower under the persistent path).
   const bool persist_ok = !no_persist && !no_cf && c->Minv != nullptr
-      && (total + 255) / 256 <= 256;
+      && c->N <= 1500 && (total + 255) / 256 <= 256;
   if (persist_ok && persist_full) {
     launch_solve_persist(c, X, nbr, tol, Delta0, accept_rho, s, 1);
     fence_signal(c, F_SOLVE_OUT, s);
