@@ -72,6 +72,7 @@ enum CtrlSlot {
   C_GN1SQ = 20,
   C_RHO = 21,
   C_HLEN = 22,     // valid Krylov-history entries (radius-independent)
+  C_SHRINKS = 23,  // radius-shrink attempts taken (batched shrink loop)
   C_DOT0 = 24,   // scratch dot slots (cleared by ctrl kernels)
   C_DOT1 = 25,
   C_DOT2 = 26,
@@ -1694,6 +1695,7 @@ __global__ void k_ctrl_shrink(double* ctrl) {
   if (threadIdx.x != 0) return;
   if (ctrl[C_STATUS] != (double)ST_TCG_STOP) return;
   ctrl[C_RADIUS] *= 0.25;
+  ctrl[C_SHRINKS] += 1.0;
 }
 
 // ---------------------------------------------------------------------
@@ -2585,19 +2587,22 @@ static bool solve_presync(DpoCtx* c, double* X, const double* nbr,
   return true;
 }
 
-// Post-sync part of the solve: acceptance decision + (rare) shrink loop.
-// Assumes the pre-sync sequence has completed on stream s and ctrl_host
-// holds the control block.
+// Post-sync part of the solve: acceptance decision + (rare) shrink
+// loop. Assumes the pre-sync sequence has completed on stream s and
+// ctrl_host holds the control block. A rejected first candidate
+// enqueues the ENTIRE guarded radius-shrink loop in one batch (each
+// attempt's kernels no-op via the ctrl guard once a candidate is
+// accepted) behind a single stream sync, instead of a host round-trip
+// per attempt.
 static int solve_postsync(DpoCtx* c, double* X, double accept_rho,
                           int max_shrink, int compute_final_gn,
                           double* stats_out, hipStream_t s) {
   const int n = c->n, d = c->d, r = c->r;
   const long total = c->total;
   const int gvec = (int)((total + 255) / 256);
-  int status = ST_GIVE_UP;
-  int shrinks = 0;
-  for (int attempt = 0; attempt <= max_shrink; ++attempt) {
-    if (attempt > 0) {
+  int st = (int)c->ctrl_host[C_STATUS];
+  if (st == ST_TCG_STOP && max_shrink > 0) {
+    for (int attempt = 1; attempt <= max_shrink; ++attempt) {
       hipLaunchKernelGGL(k_ctrl_shrink, dim3(1), dim3(64), 0, s, c->ctrl);
       hipLaunchKernelGGL(k_ctrl_candidate, dim3(1), dim3(64), 0, s,
                          c->ctrl);
@@ -2611,20 +2616,19 @@ static int solve_postsync(DpoCtx* c, double* X, double accept_rho,
                            C_DOT0, C_DOT2, ST_TCG_STOP, s);
       hipLaunchKernelGGL(k_ctrl_accept, dim3(1), dim3(64), 0, s, c->ctrl,
                          accept_rho);
-      DPO_CHECK(hipMemcpyAsync(c->ctrl_host, c->ctrl,
-                               CTRL_SIZE * sizeof(double),
-                               hipMemcpyDeviceToHost, s));
-      DPO_CHECK(hipStreamSynchronize(s));
     }
-    int st = (int)c->ctrl_host[C_STATUS];
-    if (st == ST_ACCEPTED) {
-      DPO_CHECK(hipMemcpyAsync(X, c->Xprop, total * sizeof(double),
-                               hipMemcpyDeviceToDevice, s));
-      status = st;
-      break;
-    }
-    if (st == ST_NO_UPDATE) { status = st; break; }
-    shrinks++;
+    hipLaunchKernelGGL(k_ctrl_to_host, dim3(1), dim3(64), 0, s, c->ctrl,
+                       c->ctrl_host_dev, CTRL_SIZE);
+    DPO_CHECK(hipStreamSynchronize(s));
+    st = (int)c->ctrl_host[C_STATUS];
+  }
+  const int shrinks = (int)c->ctrl_host[C_SHRINKS];
+  int status = st;
+  if (st == ST_ACCEPTED) {
+    DPO_CHECK(hipMemcpyAsync(X, c->Xprop, total * sizeof(double),
+                             hipMemcpyDeviceToDevice, s));
+  } else if (st == ST_TCG_STOP) {
+    status = ST_GIVE_UP;  // every radius rejected: keep the iterate
   }
 
   double f_init = c->ctrl_host[C_FX];
