@@ -532,12 +532,20 @@ class DistributedRBCDDriver:
         single hipGraph via torch.cuda.graph and replayed per round;
         re-captured when an agent's problem pointers change (GNC Q
         rebuild)."""
-        import os as _os
         import torch
+        modes = getattr(self, "_eval_env_modes", None)
+        if modes is None:
+            # env decisions are fixed per driver (read once, not per round)
+            import os as _os
+            _sync_eval_default = "0" if self.selection == "colored" else "1"
+            modes = self._eval_env_modes = (
+                _os.environ.get("DPO_DRIVER_EVAL_GRAPH", "0") == "1",
+                _os.environ.get("DPO_SYNC_EVAL", _sync_eval_default) == "1")
+        driver_graph, sync_eval = modes
         gen = sum(getattr(a, "_packed_generation", 0)
                   for a in self.local_agents.values())
         cache = getattr(self, "_eval_graph", None)
-        if _os.environ.get("DPO_DRIVER_EVAL_GRAPH", "0") != "1":
+        if not driver_graph:
             # Default: per-agent hip-captured eval graphs (or the colored
             # schedule's per-stream fan-out). A single driver-level
             # torch-captured graph over all agents measured ~20% faster
@@ -546,8 +554,7 @@ class DistributedRBCDDriver:
             # iteration-count scatter — parity with the reference's
             # deterministic trajectories wins. Opt in with
             # DPO_DRIVER_EVAL_GRAPH=1.
-            _sync_eval_default = "0" if self.selection == "colored" else "1"
-            if _os.environ.get("DPO_SYNC_EVAL", _sync_eval_default) == "1":
+            if sync_eval:
                 evalmat.zero_()
                 for rb, a in self.local_agents.items():
                     a._packed_eval(out=None)
