@@ -156,3 +156,19 @@ def test_logger_roundtrip(tmp_path):
     assert all(abs(m.weight - 0.5) < 1e-12 for m in back)
     back2 = lg.load_measurements("meas.csv", load_weights=False)
     assert all(m.weight == 1.0 for m in back2)
+
+
+def test_inner_tol_propagates_to_agents():
+    """DistributedRBCDDriver(inner_tol=...) must reach every agent's
+    local trust-region solver (bench.py relies on inner_tol=0 for
+    state-independent per-round work)."""
+    from dpo_amd.comm import Comm
+    from dpo_amd.dist_driver import DistributedRBCDDriver
+    from dpo_amd.synthetic import grid3d
+    meas, n = grid3d(side=3, seed=2)
+    drv = DistributedRBCDDriver(meas, n, 2, Comm(), r=5,
+                                partition="contiguous", inner_tol=0.25)
+    for a in drv.local_agents.values():
+        assert a.params.inner_tol == 0.25
+    res = drv.run(max_iters=3, gradnorm_tol=0.0)
+    assert res.iterations == 3
