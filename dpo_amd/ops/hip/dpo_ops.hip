@@ -761,6 +761,9 @@ __global__ void k_precond_jacobi(const double* __restrict__ L,
 //   snapshot: eta_snap[j] = eta, delta_snap[j] = delta  (pre-update)
 //   eta += coef * delta;  if (!stop_pending) { r += alpha*Hd; rr+=r^2 }
 // ---------------------------------------------------------------------
+// Two elements per thread: the kernel is a pure fp64 stream (4 reads
+// + 4 writes per element) and was memory-latency-bound at 1M-pose
+// scale with one 8-byte access per lane per array.
 template <int CF = CF_NONE>
 __global__ void k_tcg_update(double* __restrict__ eta,
                              double* __restrict__ rvec,
@@ -771,26 +774,30 @@ __global__ void k_tcg_update(double* __restrict__ eta,
                              double* __restrict__ ctrl,
                              long total) {
   if (guarded_off(ctrl, ST_RUN)) return;
-  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long i0 = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 2;
   const double coef = ctrl[C_COEF];
   const int stop_pending = (int)ctrl[C_STOP_PENDING];
   const int j = (int)ctrl[C_ITER];
   double rr = 0.0;
-  if (i < total) {
-    const double dl = delta[i];
-    // j == 0: eta starts the solve at 0 — write instead of accumulate,
-    // so the eta buffer needs no per-solve zeroing pass
-    const double et = (j == 0) ? 0.0 : eta[i];
-    // snapshots are write-once, re-read only on a (rare) rejection
-    // replay: nontemporal stores skip L2 write-allocate, halving the
-    // kernel's cache pressure (dominant at 1M-pose scale)
-    __builtin_nontemporal_store(et, &eta_snap[(size_t)j * total + i]);
-    __builtin_nontemporal_store(dl, &delta_snap[(size_t)j * total + i]);
-    eta[i] = fma(coef, dl, et);
-    if (!stop_pending) {
-      const double rn = fma(coef, Hd[i], rvec[i]);
-      rvec[i] = rn;
-      rr = rn * rn;
+  #pragma unroll
+  for (int u = 0; u < 2; ++u) {
+    const long i = i0 + u;
+    if (i < total) {
+      const double dl = delta[i];
+      // j == 0: eta starts the solve at 0 — write instead of
+      // accumulate, so the eta buffer needs no per-solve zeroing pass
+      const double et = (j == 0) ? 0.0 : eta[i];
+      // snapshots are write-once, re-read only on a (rare) rejection
+      // replay: nontemporal stores skip L2 write-allocate, halving
+      // the kernel's cache pressure (dominant at 1M-pose scale)
+      __builtin_nontemporal_store(et, &eta_snap[(size_t)j * total + i]);
+      __builtin_nontemporal_store(dl, &delta_snap[(size_t)j * total + i]);
+      eta[i] = fma(coef, dl, et);
+      if (!stop_pending) {
+        const double rn = fma(coef, Hd[i], rvec[i]);
+        rvec[i] = rn;
+        rr = fma(rn, rn, rr);
+      }
     }
   }
   if (!stop_pending) block_reduce_atomic(rr, ctrl + C_DOT1);
@@ -2000,7 +2007,8 @@ void dpo_tcg_update(double* eta, double* rvec, const double* delta,
                     const double* Hd, double* eta_snap, double* delta_snap,
                     double* ctrl, long total, void* stream) {
   hipStream_t s = (hipStream_t)stream;
-  hipLaunchKernelGGL((k_tcg_update<CF_NONE>), dim3(blocks_for(total, 256)),
+  hipLaunchKernelGGL((k_tcg_update<CF_NONE>),
+                     dim3(blocks_for((total + 1) / 2, 256)),
                      dim3(256), 0, s, eta, rvec, delta, Hd, eta_snap,
                      delta_snap, ctrl, total);
 }
@@ -2482,17 +2490,21 @@ static void enqueue_solve_body(DpoCtx* c, double* X, const double* nbr,
                            nullptr, c->Hd, c->delta, c->ctrl, n, d, r,
                            C_DOT0, -1, ST_RUN, s);
       hipLaunchKernelGGL(k_ctrl_alpha, dim3(1), dim3(64), 0, s, c->ctrl);
-      hipLaunchKernelGGL((k_tcg_update<CF_NONE>), dim3(gvec), dim3(256),
-                         0, s, c->eta, c->rvec, c->delta, c->Hd,
-                         c->eta_snap, c->delta_snap, c->ctrl, total);
+      hipLaunchKernelGGL((k_tcg_update<CF_NONE>),
+                         dim3(blocks_for((total + 1) / 2, 256)),
+                         dim3(256), 0, s, c->eta, c->rvec, c->delta,
+                         c->Hd, c->eta_snap, c->delta_snap, c->ctrl,
+                         total);
       hipLaunchKernelGGL(k_ctrl_rr, dim3(1), dim3(64), 0, s, c->ctrl);
     } else {
       launch_hess_fused<0, CF_ALPHA>(
           c->q_rp, c->q_ci, c->q_vals, c->delta, X, nullptr, c->Hd,
           c->delta, c->ctrl, n, d, r, C_DOT0, -1, ST_RUN, s);
-      hipLaunchKernelGGL((k_tcg_update<CF_RR>), dim3(gvec), dim3(256),
-                         0, s, c->eta, c->rvec, c->delta, c->Hd,
-                         c->eta_snap, c->delta_snap, c->ctrl, total);
+      hipLaunchKernelGGL((k_tcg_update<CF_RR>),
+                         dim3(blocks_for((total + 1) / 2, 256)),
+                         dim3(256), 0, s, c->eta, c->rvec, c->delta,
+                         c->Hd, c->eta_snap, c->delta_snap, c->ctrl,
+                         total);
     }
     ctx_precond(c, c->rvec, c->z, s);
     if (no_cf) {
