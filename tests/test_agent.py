@@ -218,3 +218,49 @@ def test_gradient_descent_ls_descends():
     assert f1 <= f0
     # line search should make real progress on a noisy instance
     assert f1 < f0 - 1e-6 * max(1.0, abs(f0))
+
+
+def test_agent_api_surface():
+    """Smaller public-API behaviors mirrored from the reference:
+    duplicate-measurement detection (PGOAgent.cpp:1291-1299), shared
+    measurement weight get/set (the owner-computes weight sync), and
+    global-frame pose accessors being consistent with the trajectory."""
+    meas, n = grid3d(side=2, seed=9, rot_noise=0.05, tran_noise=0.05)
+    odo = [m for m in meas if m.p1 + 1 == m.p2]
+    lc = [m for m in meas if m.p1 + 1 != m.p2]
+    # one shared loop closure to a fictional neighbor robot 1
+    shared = RelativeSEMeasurement(0, 1, 2, 0, np.eye(3),
+                                   np.zeros(3), 100.0, 100.0)
+    a = PGOAgent(0, PGOAgentParams(d=3, r=5, num_robots=2))
+    a.set_pose_graph(odo, lc, [shared])
+
+    # duplicate detection
+    assert PGOAgent.is_duplicate_measurement(shared, [shared])
+    other = RelativeSEMeasurement(0, 1, 3, 0, np.eye(3),
+                                  np.zeros(3), 100.0, 100.0)
+    assert not PGOAgent.is_duplicate_measurement(other, [shared])
+
+    # weight get/set round-trip
+    assert a.set_measurement_weight((0, 2), (1, 0), 0.25)
+    assert not a.set_measurement_weight((0, 7), (1, 0), 0.5)
+    weights = dict(((s, d), w) for s, d, w
+                   in a.get_shared_measurement_weights())
+    assert weights[((0, 2), (1, 0))] == 0.25
+
+    # global-frame accessors agree with the rounded trajectory
+    a.set_global_anchor(a.get_shared_pose(0))
+    T = a.get_trajectory_in_global_frame()
+    dh = 4
+    for p in range(a.n):
+        Tp = a.get_pose_in_global_frame(p)
+        assert np.allclose(Tp, T[:, p * dh:(p + 1) * dh], atol=1e-12)
+
+    # termination consensus: all agents INITIALIZED + ready
+    assert not a.should_terminate()
+    st = a.get_status()
+    st.ready_to_terminate = True
+    a.team_status[0] = st
+    import dataclasses
+    nb = dataclasses.replace(st, agent_id=1)
+    a.set_neighbor_status(nb)
+    assert a.should_terminate()
