@@ -372,6 +372,20 @@ class DistributedRBCDDriver:
         del flats
         fout = open(trace_file, "w") if (trace_file and
                                          self.comm.rank == 0) else None
+        # the reference checks convergence BEFORE the first iteration
+        # (MultiRobotExample.cpp:302-305): an already-converged start
+        # reports 0 iterations (e.g. kitti_08)
+        if gradnorm_tol > 0.0:
+            cost0, gn20 = self._evaluate()
+            gradnorm0 = float(np.sqrt(gn20.sum()))
+            if gradnorm0 < gradnorm_tol:
+                res.converged = True
+                res.final_cost = 2.0 * cost0
+                res.final_gradnorm = gradnorm0
+                if fout:
+                    fout.close()
+                res.elapsed_s = time.perf_counter() - t0
+                return res
         for it in range(max_iters):
             if self.selection == "colored":
                 color = it % self._num_colors
@@ -633,6 +647,22 @@ class DistributedRBCDDriver:
             self._round_counter = 0
         inner = next(iter(self.local_agents.values())).params \
             .robust_opt_inner_iters if self.local_agents else 30
+        # reference semantics: convergence is checked BEFORE the first
+        # iteration (an already-converged start reports 0 iterations)
+        if gradnorm_tol > 0.0:
+            self._packed_eval_phase(evalmat)
+            self.comm.all_reduce_sum_(evalmat)
+            ev0 = evalmat.cpu().numpy()
+            gradnorm0 = float(np.sqrt(ev0[:, 2].sum()))
+            if gradnorm0 < gradnorm_tol:
+                res.converged = True
+                res.final_cost = 2.0 * float((ev0[:, 0] - ev0[:, 1]).sum())
+                res.final_gradnorm = gradnorm0
+                if fout:
+                    fout.close()
+                res.elapsed_s = time.perf_counter() - t0
+                self._sync_anchor()
+                return res
         # Pipelined evaluation readback: when no per-round host decision
         # is needed (no greedy selection, no gradient-norm stop test),
         # the per-round eval scalars land in a device-side ring and are
