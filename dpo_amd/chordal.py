@@ -81,19 +81,74 @@ def _sparse_lsq(A: sp.spmatrix, b: np.ndarray) -> np.ndarray:
     return np.asarray(x)
 
 
+def _sparse_lsq_cgls(A: sp.spmatrix, b: np.ndarray, device: str = "cpu",
+                     tol: float = 1e-10, max_iters: int = 20000
+                     ) -> np.ndarray:
+    """min ||A x + b||_2 via CGLS (CG on the normal equations without
+    forming A^T A), with diagonal column scaling. Pure torch ops, so the
+    same code runs on CPU or a GPU — the offload path the reference's
+    SPQR cannot take. Accuracy matches the direct solve to the CG
+    tolerance (see tests/test_utils.py::test_chordal_cgls_matches_direct)."""
+    import torch
+    A = A.tocsr()
+    # column scaling: unit-norm columns precondition the normal equations
+    cn = np.sqrt(np.asarray(A.multiply(A).sum(axis=0)).ravel())
+    cn[cn == 0.0] = 1.0
+    D = sp.diags(1.0 / cn)
+    As = (A @ D).tocsr()
+    At = As.T.tocsr()
+
+    def to_torch(M):
+        return torch.sparse_csr_tensor(
+            torch.from_numpy(M.indptr.astype(np.int64)),
+            torch.from_numpy(M.indices.astype(np.int64)),
+            torch.from_numpy(M.data.astype(np.float64)),
+            size=M.shape, device=device)
+
+    tA = to_torch(As)
+    tAt = to_torch(At)
+    rhs = torch.from_numpy(np.ascontiguousarray(-b)).to(device)
+    x = torch.zeros(As.shape[1], dtype=torch.float64, device=device)
+    r = rhs.clone()
+    s = tAt @ r.unsqueeze(1)
+    p = s.clone()
+    gamma = float((s * s).sum())
+    g0 = gamma
+    for _ in range(max_iters):
+        if gamma <= tol * tol * max(g0, 1e-300):
+            break
+        q = tA @ p
+        qq = float((q * q).sum())
+        if qq == 0.0:
+            break
+        alpha = gamma / qq
+        x += alpha * p.squeeze(1)
+        r -= alpha * q.squeeze(1)
+        s = tAt @ r.unsqueeze(1)
+        g_new = float((s * s).sum())
+        p = s + (g_new / gamma) * p
+        gamma = g_new
+    return (x.cpu().numpy() / cn)
+
+
 def chordal_initialization(d: int, num_poses: int,
-                           measurements: Sequence[RelativeSEMeasurement]
-                           ) -> np.ndarray:
-    """Returns T (d, (d+1) n): [R1 t1 R2 t2 ...] with pose 0 = identity."""
+                           measurements: Sequence[RelativeSEMeasurement],
+                           method: str = "direct",
+                           device: str = "cpu") -> np.ndarray:
+    """Returns T (d, (d+1) n): [R1 t1 R2 t2 ...] with pose 0 = identity.
+    method="direct": sparse LU of the normal equations (reference-exact
+    quality); method="cgls": iterative CGLS in torch, runnable on a GPU."""
     assert measurements, "chordal initialization needs measurements"
     d2 = d * d
     B1, B2, B3 = construct_b_matrices(measurements, num_poses, d)
+    lsq = (_sparse_lsq if method == "direct"
+           else lambda A, b: _sparse_lsq_cgls(A, b, device=device))
 
     # Rotations: pin pose 0 to I, solve for the rest.
     B3red = B3[:, d2:]
     Id = np.eye(d)
     cR = B3[:, :d2] @ Id.flatten(order="F")
-    rvec = _sparse_lsq(B3red, cR)
+    rvec = lsq(B3red, cR)
     Rall = np.zeros((d, d * num_poses))
     Rall[:, :d] = Id
     Rall[:, d:] = rvec.reshape(d, d * (num_poses - 1), order="F")
@@ -102,7 +157,7 @@ def chordal_initialization(d: int, num_poses: int,
             Rall[:, i * d:(i + 1) * d])
 
     # Translations from the rounded rotations.
-    t = recover_translations(B1, B2, Rall)
+    t = recover_translations(B1, B2, Rall, method=method, device=device)
 
     T = np.zeros((d, num_poses * (d + 1)))
     for i in range(num_poses):
@@ -112,7 +167,8 @@ def chordal_initialization(d: int, num_poses: int,
 
 
 def recover_translations(B1: sp.spmatrix, B2: sp.spmatrix,
-                         R: np.ndarray) -> np.ndarray:
+                         R: np.ndarray, method: str = "direct",
+                         device: str = "cpu") -> np.ndarray:
     """Second least-squares solve for translations given rotations
     (reference DPGO_utils.cpp:434-461)."""
     d = R.shape[0]
@@ -120,7 +176,10 @@ def recover_translations(B1: sp.spmatrix, B2: sp.spmatrix,
     rvec = R.flatten(order="F")
     B1red = B1[:, d:]
     c = B2 @ rvec
-    tred = _sparse_lsq(B1red, c)
+    if method == "direct":
+        tred = _sparse_lsq(B1red, c)
+    else:
+        tred = _sparse_lsq_cgls(B1red, c, device=device)
     t = np.zeros((d, n))
     t[:, 1:] = tred.reshape(d, n - 1, order="F")
     return t
