@@ -31,6 +31,9 @@
 
 // Debug/bisection knobs: DPO_NO_SOLVE_GRAPH=1 / DPO_NO_EVAL_GRAPH=1 run
 // the solve / eval sequences eagerly instead of via cached hipGraphs.
+// teardown/cleanup calls whose errors are deliberately ignored
+static inline void dpo_ignore(hipError_t) {}
+
 static bool dpo_env_flag(const char* name) {
   const char* v = getenv(name);
   return v && v[0] == '1';
@@ -2207,8 +2210,14 @@ struct DpoCtx {
   // stable pointer keeps the eval graph cache hot)
   double* eval3 = nullptr;
   void invalidate_graphs() {
-    if (solve_graph) { hipGraphExecDestroy(solve_graph); solve_graph = nullptr; }
-    if (eval_graph) { hipGraphExecDestroy(eval_graph); eval_graph = nullptr; }
+    if (solve_graph) {
+      dpo_ignore(hipGraphExecDestroy(solve_graph));
+      solve_graph = nullptr;
+    }
+    if (eval_graph) {
+      dpo_ignore(hipGraphExecDestroy(eval_graph));
+      eval_graph = nullptr;
+    }
   }
 };
 
@@ -2339,16 +2348,16 @@ void* dpo_ctx_create(int n, int d, int r, int max_inner) {
 
 void dpo_ctx_destroy(void* h) {
   DpoCtx* c = (DpoCtx*)h;
-  hipFree(c->W); hipFree(c->grad); hipFree(c->eta); hipFree(c->delta);
-  hipFree(c->rvec); hipFree(c->z); hipFree(c->Hd); hipFree(c->step);
-  hipFree(c->Xprop); hipFree(c->eta_snap); hipFree(c->delta_snap);
+  dpo_ignore(hipFree(c->W)); dpo_ignore(hipFree(c->grad)); dpo_ignore(hipFree(c->eta)); dpo_ignore(hipFree(c->delta));
+  dpo_ignore(hipFree(c->rvec)); dpo_ignore(hipFree(c->z)); dpo_ignore(hipFree(c->Hd)); dpo_ignore(hipFree(c->step));
+  dpo_ignore(hipFree(c->Xprop)); dpo_ignore(hipFree(c->eta_snap)); dpo_ignore(hipFree(c->delta_snap));
   c->invalidate_graphs();
-  if (c->cap_stream) hipStreamDestroy(c->cap_stream);
-  if (c->exec_stream) hipStreamDestroy(c->exec_stream);
-  if (c->start_event) hipEventDestroy(c->start_event);
-  if (c->done_event) hipEventDestroy(c->done_event);
-  hipFree(c->ctrl); hipFree(c->G_buf); hipHostFree(c->ctrl_host);
-  hipFree(c->fences); hipFree(c->eval3); hipFree(c->gbar);
+  if (c->cap_stream) dpo_ignore(hipStreamDestroy(c->cap_stream));
+  if (c->exec_stream) dpo_ignore(hipStreamDestroy(c->exec_stream));
+  if (c->start_event) dpo_ignore(hipEventDestroy(c->start_event));
+  if (c->done_event) dpo_ignore(hipEventDestroy(c->done_event));
+  dpo_ignore(hipFree(c->ctrl)); dpo_ignore(hipFree(c->G_buf)); dpo_ignore(hipHostFree(c->ctrl_host));
+  dpo_ignore(hipFree(c->fences)); dpo_ignore(hipFree(c->eval3)); dpo_ignore(hipFree(c->gbar));
   delete c;
 }
 
@@ -2562,7 +2571,7 @@ static bool solve_presync(DpoCtx* c, double* X, const double* nbr,
                    && c->solve_replays < DpoCtx::max_replays();
   if (!key_match) {
     if (c->solve_graph) {
-      hipGraphExecDestroy(c->solve_graph);
+      dpo_ignore(hipGraphExecDestroy(c->solve_graph));
       c->solve_graph = nullptr;
     }
     c->solve_replays = 0;
@@ -2575,17 +2584,17 @@ static bool solve_presync(DpoCtx* c, double* X, const double* nbr,
       rc = hipStreamEndCapture(cs_, &graph);
     }
     hipStreamCaptureStatus capst = hipStreamCaptureStatusNone;
-    hipStreamIsCapturing(cs_, &capst);
+    dpo_ignore(hipStreamIsCapturing(cs_, &capst));
     if (capst != hipStreamCaptureStatusNone) {
       hipGraph_t dead = nullptr;
-      hipStreamEndCapture(cs_, &dead);
-      if (dead) hipGraphDestroy(dead);
+      dpo_ignore(hipStreamEndCapture(cs_, &dead));
+      if (dead) dpo_ignore(hipGraphDestroy(dead));
     }
     if (rc == hipSuccess && graph) {
       rc = hipGraphInstantiate(&c->solve_graph, graph, nullptr, nullptr, 0);
-      hipGraphDestroy(graph);
+      dpo_ignore(hipGraphDestroy(graph));
     }
-    hipGetLastError();  // clear sticky capture-related error state
+    dpo_ignore(hipGetLastError());  // clear sticky capture-related error state
     if (rc != hipSuccess || !c->solve_graph) {
       // capture unavailable: run eagerly on the caller's stream
       c->solve_graph = nullptr;
@@ -2843,7 +2852,7 @@ static void round_eval_impl(DpoCtx* c, const double* X, const double* nbr,
                && c->eval_replays < DpoCtx::max_replays();
   if (!match) {
     if (c->eval_graph) {
-      hipGraphExecDestroy(c->eval_graph);
+      dpo_ignore(hipGraphExecDestroy(c->eval_graph));
       c->eval_graph = nullptr;
     }
     c->eval_replays = 0;
@@ -2856,17 +2865,17 @@ static void round_eval_impl(DpoCtx* c, const double* X, const double* nbr,
       rc = hipStreamEndCapture(cs_, &graph);
     }
     hipStreamCaptureStatus capst = hipStreamCaptureStatusNone;
-    hipStreamIsCapturing(cs_, &capst);
+    dpo_ignore(hipStreamIsCapturing(cs_, &capst));
     if (capst != hipStreamCaptureStatusNone) {
       hipGraph_t dead = nullptr;
-      hipStreamEndCapture(cs_, &dead);
-      if (dead) hipGraphDestroy(dead);
+      dpo_ignore(hipStreamEndCapture(cs_, &dead));
+      if (dead) dpo_ignore(hipGraphDestroy(dead));
     }
     if (rc == hipSuccess && graph) {
       rc = hipGraphInstantiate(&c->eval_graph, graph, nullptr, nullptr, 0);
-      hipGraphDestroy(graph);
+      dpo_ignore(hipGraphDestroy(graph));
     }
-    hipGetLastError();  // clear sticky capture-related error state
+    dpo_ignore(hipGetLastError());  // clear sticky capture-related error state
     if (rc != hipSuccess || !c->eval_graph) {
       c->eval_graph = nullptr;
       enqueue_eval_body(c, X, nbr, out_dev, s);
@@ -2952,10 +2961,10 @@ struct DpoGroup {
 void dpo_group_destroy(void* g) {
   DpoGroup* gr = (DpoGroup*)g;
   if (!gr) return;
-  hipFree(gr->d_srcs);
-  hipFree(gr->d_rows);
-  hipFree(gr->d_fin);
-  hipFree(gr->d_fout);
+  dpo_ignore(hipFree(gr->d_srcs));
+  dpo_ignore(hipFree(gr->d_rows));
+  dpo_ignore(hipFree(gr->d_fin));
+  dpo_ignore(hipFree(gr->d_fout));
   free(gr->cs); free(gr->Xs); free(gr->nbrs);
   delete gr;
 }
