@@ -391,6 +391,8 @@ class DistributedRBCDDriver:
                 color = it % self._num_colors
                 active = [rb for rb in range(self.num_robots)
                           if self._colors[rb] == color]
+            elif self.selection == "round_robin":
+                active = [it % self.num_robots]
             else:
                 active = [selected]
             for rb, a in self.local_agents.items():
@@ -412,7 +414,8 @@ class DistributedRBCDDriver:
                 break
             if time_limit_s and time.perf_counter() - t0 > time_limit_s:
                 break
-            selected = int(np.argmax(gn2))
+            if self.selection != "round_robin":
+                selected = int(np.argmax(gn2))
         if res.trace:
             res.final_cost, res.final_gradnorm = res.trace[-1]
         if fout:
@@ -756,6 +759,9 @@ class DistributedRBCDDriver:
                     a.robust_cost.update()
             if self.selection == "colored":
                 active = color_active[it % self._num_colors]
+            elif self.selection == "round_robin":
+                selected = it % self.num_robots
+                active = [selected]
             else:
                 active = [selected]
             if accel:
@@ -846,7 +852,8 @@ class DistributedRBCDDriver:
                 break
             if time_limit_s and time.perf_counter() - t0 > time_limit_s:
                 break
-            selected = int(np.argmax(gn2))
+            if self.selection != "round_robin":
+                selected = int(np.argmax(gn2))
         if chunk > 1:
             flush_ring()   # safety net: all exits above should have flushed
         if _timing and self.comm.rank == 0 and res.iterations:

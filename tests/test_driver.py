@@ -172,3 +172,21 @@ def test_inner_tol_propagates_to_agents():
         assert a.params.inner_tol == 0.25
     res = drv.run(max_iters=3, gradnorm_tol=0.0)
     assert res.iterations == 3
+
+
+def test_dist_driver_round_robin():
+    """Cyclic (round-robin) agent selection in the distributed driver."""
+    from dpo_amd.comm import Comm
+    from dpo_amd.dist_driver import DistributedRBCDDriver
+    meas, n = grid3d(side=3, seed=4, rot_noise=0.1, tran_noise=0.05)
+    drv = DistributedRBCDDriver(meas, n, 3, Comm(), r=5,
+                                partition="contiguous",
+                                selection="round_robin")
+    res = drv.run(max_iters=60)
+    assert res.converged
+    # still reaches the same optimum as greedy
+    greedy = DistributedRBCDDriver(meas, n, 3, Comm(), r=5,
+                                   partition="contiguous")
+    res_g = greedy.run(max_iters=200)
+    assert abs(res.final_cost - res_g.final_cost) < 1e-3 * max(
+        1.0, abs(res_g.final_cost))
