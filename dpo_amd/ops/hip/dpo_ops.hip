@@ -13,10 +13,11 @@
 //    Stiefel block, the last the translation. Consecutive lanes read
 //    consecutive (row, col) elements -> coalesced.
 //  * Q is (d+1)x(d+1) block-CSR ("BSR"): row_ptr (n+1), col_idx (nnzb),
-//    vals (nnzb, dh, dh). The SpMM computes a whole pose-block row per
-//    THREAD (the dh x r accumulator tile lives in registers, each Q
-//    block is one or two cache lines); fp64 FMA; bandwidth/latency-
-//    bound at these sizes.
+//    vals (nnzb, dh, dh). The standalone SpMM assigns a pose-block row
+//    per dh*r-thread group (one output element per thread); the FUSED
+//    Hessian/projection kernel computes a whole pose-block row per
+//    thread with the dh x r accumulator tile in registers. fp64 FMA;
+//    bandwidth/latency-bound at these sizes.
 //  * Wavefront = 64; blocks are multiples of 64 threads.
 //
 // Functional parity targets: Hessian-vec V*Q
