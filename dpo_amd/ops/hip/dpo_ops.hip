@@ -2738,7 +2738,8 @@ int dpo_round_solve_finish(void* h, int max_shrink, double* stats_out,
                            void* join_stream) {
   DpoCtx* c = (DpoCtx*)h;
   DPO_CHECK(hipStreamSynchronize(c->exec_stream));
-  if (dpo_env_flag("DPO_DBG_PERSIST")) {
+  static const bool dbg_persist = dpo_env_flag("DPO_DBG_PERSIST");
+  if (dbg_persist) {
     const double* ts = c->ctrl_host + CTRL_SIZE;
     fprintf(stderr, "[persist us] zeroG %.1f grad %.1f z0 %.1f loop %.1f"
             " end %.1f cand %.1f pub %.1f\n",
