@@ -338,6 +338,103 @@ void connectivity_fixup(const Graph& g, int k, double max_w,
   }
 }
 
+// Coarsening that only matches vertices within the same class, so an
+// existing partition projects losslessly onto the coarse graph
+// (iterated-multilevel V-cycles, the main quality lever of modern
+// multilevel partitioners).
+Graph coarsen_respecting(const Graph& g, const std::vector<int>& part,
+                         std::vector<int>& cmap, std::mt19937& rng) {
+  const int n = g.n;
+  std::vector<int> match(n, -1);
+  std::vector<int> order(n);
+  std::iota(order.begin(), order.end(), 0);
+  std::shuffle(order.begin(), order.end(), rng);
+  for (int u : order) {
+    if (match[u] >= 0) continue;
+    int best = -1;
+    double bw = -1.0;
+    for (int e = g.xadj[u]; e < g.xadj[u + 1]; ++e) {
+      int v = g.adjncy[e];
+      if (match[v] < 0 && v != u && part[v] == part[u] && g.adjw[e] > bw) {
+        bw = g.adjw[e];
+        best = v;
+      }
+    }
+    match[u] = (best >= 0) ? best : u;
+    if (best >= 0) match[best] = u;
+  }
+  cmap.assign(n, -1);
+  int nc = 0;
+  for (int u = 0; u < n; ++u)
+    if (cmap[u] < 0) {
+      cmap[u] = nc;
+      cmap[match[u]] = nc;
+      ++nc;
+    }
+  Graph c;
+  c.n = nc;
+  c.vwgt.assign(nc, 0.0);
+  for (int u = 0; u < n; ++u) c.vwgt[cmap[u]] += g.vwgt[u];
+  std::vector<std::vector<std::pair<int, double>>> tmp(nc);
+  std::vector<int> seen(nc, -1);
+  std::vector<int> pos(nc, 0);
+  for (int u = 0; u < n; ++u) {
+    int cu = cmap[u];
+    for (int e = g.xadj[u]; e < g.xadj[u + 1]; ++e) {
+      int cv = cmap[g.adjncy[e]];
+      if (cu == cv) continue;
+      if (seen[cv] == cu) {
+        tmp[cu][pos[cv]].second += g.adjw[e];
+      } else {
+        seen[cv] = cu;
+        pos[cv] = (int)tmp[cu].size();
+        tmp[cu].push_back({cv, g.adjw[e]});
+      }
+    }
+  }
+  c.xadj.assign(nc + 1, 0);
+  for (int u = 0; u < nc; ++u) c.xadj[u + 1] = c.xadj[u] + (int)tmp[u].size();
+  c.adjncy.resize(c.xadj[nc]);
+  c.adjw.resize(c.xadj[nc]);
+  for (int u = 0; u < nc; ++u)
+    for (size_t j = 0; j < tmp[u].size(); ++j) {
+      c.adjncy[c.xadj[u] + j] = tmp[u][j].first;
+      c.adjw[c.xadj[u] + j] = tmp[u][j].second;
+    }
+  return c;
+}
+
+// One V-cycle: coarsen respecting `part`, refine coarse-to-fine.
+std::vector<int> vcycle(const Graph& g0, int k, double max_w,
+                        std::vector<int> part, int target,
+                        std::mt19937& rng) {
+  std::vector<Level> levels;
+  Graph g = g0;
+  std::vector<int> cur = std::move(part);
+  while (g.n > target) {
+    Level lv;
+    Graph c = coarsen_respecting(g, cur, lv.cmap, rng);
+    if (c.n >= (int)(g.n * 0.95)) break;
+    std::vector<int> cpart(c.n);
+    for (int u = 0; u < g.n; ++u) cpart[lv.cmap[u]] = cur[u];
+    lv.g = std::move(g);
+    g = std::move(c);
+    cur = std::move(cpart);
+    levels.push_back(std::move(lv));
+  }
+  refine(g, k, max_w, cur);
+  while (!levels.empty()) {
+    Level lv = std::move(levels.back());
+    levels.pop_back();
+    std::vector<int> fine(lv.g.n);
+    for (int u = 0; u < lv.g.n; ++u) fine[u] = cur[lv.cmap[u]];
+    cur = std::move(fine);
+    refine(lv.g, k, max_w, cur);
+    g = std::move(lv.g);
+  }
+  return cur;
+}
+
 std::vector<int> multilevel_once(const Graph& g0, int k, double imbalance,
                                  unsigned seed) {
   std::mt19937 rng(seed);
@@ -370,6 +467,73 @@ std::vector<int> multilevel_once(const Graph& g0, int k, double imbalance,
   }
   connectivity_fixup(g, k, max_w, part);
   refine(g, k, max_w, part);
+  // iterated V-cycles: re-coarsen respecting the current partition and
+  // re-refine at every level; keep strict improvements
+  double cut = cut_of(g, part);
+  for (int vc = 0; vc < 4; ++vc) {
+    auto cand = vcycle(g, k, max_w, part, target, rng);
+    connectivity_fixup(g, k, max_w, cand);
+    refine(g, k, max_w, cand);
+    double c2 = cut_of(g, cand);
+    if (c2 < cut - 1e-12) {
+      part = std::move(cand);
+      cut = c2;
+    } else {
+      break;
+    }
+  }
+  // iterated local search: kick a random boundary blob into the
+  // adjacent class, re-refine (one V-cycle), keep improvements — moves
+  // FM out of local minima the gradient-like passes cannot escape
+  {
+    std::vector<double> wgt(k, 0.0);
+    const int blob_cap = std::max(8, g.n / (16 * k));
+    std::uniform_int_distribution<int> un(0, g.n - 1);
+    for (int kick = 0; kick < 8; ++kick) {
+      auto cand = part;
+      for (int q = 0; q < k; ++q) wgt[q] = 0.0;
+      for (int u = 0; u < g.n; ++u) wgt[cand[u]] += g.vwgt[u];
+      int u0 = -1, tp = -1;
+      for (int tries = 0; tries < 512 && u0 < 0; ++tries) {
+        int x = un(rng);
+        for (int e = g.xadj[x]; e < g.xadj[x + 1]; ++e)
+          if (cand[g.adjncy[e]] != cand[x]) {
+            u0 = x;
+            tp = cand[g.adjncy[e]];
+            break;
+          }
+      }
+      if (u0 < 0) break;
+      // grow a same-class BFS blob around u0 and flip it to tp
+      const int p0 = cand[u0];
+      std::vector<int> blob{u0}, q{u0};
+      std::vector<char> inblob(g.n, 0);
+      inblob[u0] = 1;
+      double bw = g.vwgt[u0];
+      for (size_t h = 0; h < q.size() && (int)blob.size() < blob_cap; ++h)
+        for (int e = g.xadj[q[h]]; e < g.xadj[q[h] + 1]; ++e) {
+          int v = g.adjncy[e];
+          if (!inblob[v] && cand[v] == p0 &&
+              (int)blob.size() < blob_cap &&
+              wgt[tp] + bw + g.vwgt[v] <= max_w) {
+            inblob[v] = 1;
+            blob.push_back(v);
+            q.push_back(v);
+            bw += g.vwgt[v];
+          }
+        }
+      if (wgt[tp] + bw > max_w) continue;
+      for (int v : blob) cand[v] = tp;
+      cand = vcycle(g, k, max_w, std::move(cand), target, rng);
+      connectivity_fixup(g, k, max_w, cand);
+      refine(g, k, max_w, cand);
+      double c2 = cut_of(g, cand);
+      if (c2 < cut - 1e-12) {
+        part = std::move(cand);
+        cut = c2;
+      }
+    }
+  }
   return part;
 }
 
