@@ -219,11 +219,6 @@ class PGOAgent:
         """Build the linear term from cached neighbor poses. Returns False
         (skip update) when any required pose is missing
         (PGOAgent.cpp:806-814)."""
-        if getattr(self, "_soa", False):
-            raise NotImplementedError(
-                "SoA agents (set_pose_graph_arrays) run through the packed "
-                "GPU path; the PoseDict exchange path needs object-mode "
-                "set_pose_graph")
         n_slots = len(self._nbr_slot_order)
         buf = np.zeros((n_slots, self.dh, self.r))
         for k, pid in enumerate(self._nbr_slot_order):
@@ -233,8 +228,11 @@ class PGOAgent:
                     print(f"agent {self.id}: missing neighbor pose {pid}")
                 return False
             buf[k] = v.T  # (r, dh) -> (dh, r)
-        w = torch.tensor([m.weight for m in self.shared_lc],
-                         dtype=torch.float64)
+        if getattr(self, "_soa", False):
+            w = torch.from_numpy(self._shared_ma.weight.copy())
+        else:
+            w = torch.tensor([m.weight for m in self.shared_lc],
+                             dtype=torch.float64)
         nbr = torch.from_numpy(buf).to(self.device)
         Gt = self._g_assembler.assemble(nbr, w, self.r)
         self.problem.set_g(Gt)
@@ -346,6 +344,13 @@ class PGOAgent:
 
     def get_neighbors(self) -> List[int]:
         return sorted(self.neighbor_robot_ids)
+
+    def has_shared_lc(self) -> bool:
+        """True when this agent has inter-robot loop closures (works for
+        both object-mode and SoA agents)."""
+        if getattr(self, "_soa", False):
+            return len(self._shared_ma) > 0
+        return bool(self.shared_lc)
 
     def get_neighbor_public_poses(self, neighbor_id: int) -> List[int]:
         assert neighbor_id in self.neighbor_robot_ids
@@ -557,9 +562,9 @@ class PGOAgent:
 
         pose_dict = (self.neighbor_aux_pose_dict if acceleration
                      else self.neighbor_pose_dict)
-        if self.shared_lc and not self._construct_g(pose_dict):
+        if self.has_shared_lc() and not self._construct_g(pose_dict):
             return False
-        if not self.shared_lc:
+        if not self.has_shared_lc():
             self.problem.set_g(torch.zeros(
                 self.dh * self.n, self.r, dtype=torch.float64,
                 device=self.device))
@@ -646,6 +651,10 @@ class PGOAgent:
 
     def update_loop_closures_weights(self) -> None:
         assert self.state == PGOAgentState.INITIALIZED
+        if getattr(self, "_soa", False):
+            raise RuntimeError(
+                "SoA agents update GNC weights via the packed GPU path "
+                "(_packed_update_weights); robust SoA needs a cuda device")
         for m in self.private_lc:
             if m.is_known_inlier:
                 continue

@@ -189,15 +189,26 @@ def odometry_initialization(d: int, num_poses: int,
                             odometry: Sequence[RelativeSEMeasurement]
                             ) -> np.ndarray:
     """Dead-reckoning T_{i+1} = T_i * m_i from identity
-    (reference DPGO_utils.cpp:411-432)."""
+    (reference DPGO_utils.cpp:411-432). Gap-tolerant: a robot built from
+    a non-contiguous (multilevel) partition may be missing odometry
+    steps between locally consecutive poses — those gaps propagate with
+    an identity step (matching odometry_initialization_array)."""
     dh = d + 1
+    step = {}
+    for m in odometry:
+        assert m.p2 == m.p1 + 1, "odometry edges must link consecutive poses"
+        step[m.p1] = m
     T = np.zeros((d, num_poses * dh))
     T[:, 0:d] = np.eye(d)
-    for k, m in enumerate(odometry):
-        src, dst = k, k + 1
-        assert m.p1 == src and m.p2 == dst, "odometry must be a chain"
+    for src in range(num_poses - 1):
+        dst = src + 1
         Rsrc = T[:, src * dh:src * dh + d]
         tsrc = T[:, src * dh + d]
-        T[:, dst * dh:dst * dh + d] = Rsrc @ m.R
-        T[:, dst * dh + d] = tsrc + Rsrc @ m.t
+        m = step.get(src)
+        if m is None:
+            T[:, dst * dh:dst * dh + d] = Rsrc
+            T[:, dst * dh + d] = tsrc
+        else:
+            T[:, dst * dh:dst * dh + d] = Rsrc @ m.R
+            T[:, dst * dh + d] = tsrc + Rsrc @ m.t
     return T

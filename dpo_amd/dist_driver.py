@@ -70,6 +70,13 @@ class DistributedRBCDDriver:
         self.device = device
         from .measurements import MeasurementArray
         self._soa = isinstance(measurements, MeasurementArray)
+        if (self._soa and robust != RobustCostType.L2
+                and not str(device).startswith("cuda")):
+            raise ValueError(
+                "robust (non-L2) SoA driver needs a cuda device: GNC "
+                "weight updates run in the packed GPU path "
+                "(_packed_update_weights); on CPU pass object-mode "
+                "measurements (list of RelativeSEMeasurement)")
         d = measurements.d if self._soa else measurements[0].d
         self.d, self.r, self.n_global = d, r, num_poses
         self.dh = d + 1
@@ -119,6 +126,36 @@ class DistributedRBCDDriver:
             (odometry, private_lc, shared_lc, self.pose_map,
              self.pose_to_index, self.pose_counts) = partition_measurements(
                 measurements, num_poses, part, num_robots)
+            # global cross-edge enumeration (GNC weight sync on the dict
+            # path; owner-computes rule — PGOAgent.cpp:1201-1244): each
+            # inter-robot loop closure gets one global slot, keyed by its
+            # (global src, global dst) pose pair in input order.
+            partv = np.asarray(part, dtype=np.int64)
+            cross_fifo: Dict[Tuple[int, int], List[int]] = {}
+            n_cross = 0
+            for m in measurements:
+                if partv[m.p1] != partv[m.p2]:
+                    cross_fifo.setdefault((m.p1, m.p2), []).append(n_cross)
+                    n_cross += 1
+            self._n_cross = n_cross
+            # per-robot: shared_lc edge k -> global cross slot, owned mask
+            self._cross_map = []
+            self._cross_owned = []
+            taken: Dict[Tuple[int, int], int] = {}
+            for rb in range(num_robots):
+                idxs = []
+                owned = []
+                seen: Dict[Tuple[int, int], int] = {}
+                for m in shared_lc[rb]:
+                    g1 = self.pose_to_index[(m.r1, m.p1)]
+                    g2 = self.pose_to_index[(m.r2, m.p2)]
+                    k = seen.get((g1, g2), 0)
+                    seen[(g1, g2)] = k + 1
+                    idxs.append(cross_fifo[(g1, g2)][k])
+                    other = m.r2 if m.r1 == rb else m.r1
+                    owned.append(other > rb)
+                self._cross_map.append(np.asarray(idxs, dtype=np.int64))
+                self._cross_owned.append(np.asarray(owned, dtype=bool))
 
         # ---- agents owned by this rank ---------------------------------
         self.owner = [a % comm.world_size for a in range(num_robots)]
@@ -288,7 +325,7 @@ class DistributedRBCDDriver:
         if a.state != PGOAgentState.INITIALIZED or a.problem is None:
             return np.zeros(3)
         ok = True
-        if a.shared_lc:
+        if a.has_shared_lc():
             ok = a._construct_g(a.neighbor_pose_dict)
         if not ok:
             return np.zeros(3)
@@ -339,6 +376,27 @@ class DistributedRBCDDriver:
             if anchor is not None and np.any(anchor):
                 a.set_global_anchor(np.ascontiguousarray(anchor.T))
         return statuses
+
+    def _dict_sync_weights(self) -> None:
+        """All ranks: collect owner-computed GNC weights of inter-robot
+        loop closures and apply them on both co-owners (one all-reduce of
+        n_cross doubles; mirrors the packed path's _w_exch)."""
+        w = torch.zeros(self._n_cross, dtype=torch.float64)
+        for rb, a in self.local_agents.items():
+            cm, own = self._cross_map[rb], self._cross_owned[rb]
+            if not len(cm):
+                continue
+            ws = np.array([m.weight for m in a.shared_lc])
+            sel = np.nonzero(own)[0]
+            if len(sel):
+                w[torch.from_numpy(cm[sel])] = torch.from_numpy(ws[sel])
+        self.comm.all_reduce_sum_(w)
+        wn = w.numpy()
+        for rb, a in self.local_agents.items():
+            cm = self._cross_map[rb]
+            for k, m in enumerate(a.shared_lc):
+                m.weight = float(wn[cm[k]])
+            a.publish_weights_requested = False
 
     def _evaluate(self) -> Tuple[float, np.ndarray]:
         """Centralized cost + per-agent centralized block grad-norm^2,
@@ -397,6 +455,9 @@ class DistributedRBCDDriver:
                 active = [selected]
             for rb, a in self.local_agents.items():
                 a.iterate(rb in active)
+            if self.robust != RobustCostType.L2 and not self._soa \
+                    and self._n_cross:
+                self._dict_sync_weights()
             flats = self.comm.all_gather_flat(self._pack_rank_payload(),
                                               self.rank_sizes)
             self._unpack_and_update(flats)

@@ -191,6 +191,22 @@ class MultiRobotDriver:
                 sel.set_neighbor_status(a.get_status())
                 sel.update_aux_neighbor_poses(a.id, aux)
 
+    def _sync_shared_weights(self) -> None:
+        """Propagate GNC weights of shared loop closures from the owning
+        agent (the lower-id endpoint, which is the one that computes
+        them under the owner-computes rule, PGOAgent.cpp:1201-1244) to
+        the co-owner, so both agents optimize the same objective.
+        The reference does this through its ROS publish step; in-process
+        we apply the owner's weights directly after each round."""
+        for a in self.agents:
+            if not a.publish_weights_requested:
+                continue
+            for src, dst, w in a.get_shared_measurement_weights():
+                other = dst[0] if src[0] == a.id else src[0]
+                if other > a.id:  # a owns this edge -> push to co-owner
+                    self.agents[other].set_measurement_weight(src, dst, w)
+            a.publish_weights_requested = False
+
     def run(self, max_iters: int = 1000, gradnorm_tol: float = 0.1,
             trace_file: Optional[str] = None) -> RBCDResult:
         res = RBCDResult()
@@ -229,6 +245,7 @@ class MultiRobotDriver:
                     self._exchange_with(selected, acceleration)
                     self.agents[selected].iterate(True)
 
+                self._sync_shared_weights()
                 X = self._gather_global_x()
                 rgrad = self.central.rie_grad(X)
                 gradnorm = float(torch.linalg.norm(rgrad))

@@ -190,3 +190,82 @@ def test_dist_driver_round_robin():
     res_g = greedy.run(max_iters=200)
     assert abs(res.final_cost - res_g.final_cost) < 1e-3 * max(
         1.0, abs(res_g.final_cost))
+
+
+def test_gnc_shared_weights_consistent_between_agents():
+    """Regression (round-1 advisor, high): owner-computed GNC weights of
+    shared loop closures must be propagated to the co-owning agent; stale
+    copies silently optimize inconsistent objectives."""
+    from dpo_amd.types import RobustCostParams
+    meas, n = grid3d(side=4, seed=7, outlier_prob=0.2)
+    rp = RobustCostParams(gnc_mu_step=2.5, gnc_init_mu=1e-3)
+    drv = MultiRobotDriver(meas, n, 2, r=5, partition="contiguous",
+                           robust=RobustCostType.GNC_TLS,
+                           robust_params=rp, robust_inner_iters=10)
+    drv.run(max_iters=120, gradnorm_tol=0.1)
+    wmap = {}
+    updated = 0
+    for a in drv.agents:
+        for m in a.shared_lc:
+            key = ((m.r1, m.p1), (m.r2, m.p2))
+            if key in wmap:
+                assert abs(wmap[key] - m.weight) < 1e-12, \
+                    f"shared weight copies disagree on {key}"
+            else:
+                wmap[key] = m.weight
+            if m.weight != 1.0:
+                updated += 1
+    assert wmap, "fixture must produce shared loop closures"
+    assert updated > 0, "GNC must have updated at least one shared weight"
+
+
+def test_robust_dist_multilevel_constructs_and_runs():
+    """Regression (round-1 advisor, medium): robust dict-path driver with
+    a non-contiguous multilevel partition used to crash in
+    odometry_initialization on odometry-chain gaps."""
+    from dpo_amd.comm import Comm
+    from dpo_amd.dist_driver import DistributedRBCDDriver
+    meas, n = grid3d(side=4, seed=3, outlier_prob=0.1)
+    drv = DistributedRBCDDriver(meas, n, 3, Comm(), r=5,
+                                partition="multilevel",
+                                robust=RobustCostType.GNC_TLS)
+    res = drv.run(max_iters=40, gradnorm_tol=0.0)
+    assert res.iterations == 40
+    assert np.isfinite(res.final_cost)
+
+
+def test_odometry_initialization_gap_tolerant():
+    from dpo_amd.chordal import odometry_initialization
+    meas, n = grid3d(side=3, seed=5, rot_noise=0.05)
+    odo = [m for m in meas if m.p1 + 1 == m.p2]
+    # knock out an interior step -> identity propagation across the gap
+    odo_gap = [m for m in odo if m.p1 != 4]
+    T = odometry_initialization(3, n, odo_gap)
+    assert np.all(np.isfinite(T))
+    # pose 5 equals pose 4 (identity step over the gap)
+    dh = 4
+    assert np.allclose(T[:, 4 * dh:5 * dh], T[:, 5 * dh:6 * dh])
+
+
+def test_soa_cpu_dict_path_runs():
+    """Regression (VERDICT weak #3): a SoA driver on CPU used to fall
+    into the dict path and die in _construct_g."""
+    from dpo_amd.comm import Comm
+    from dpo_amd.dist_driver import DistributedRBCDDriver
+    from dpo_amd.synthetic import grid3d_soa
+    ma, n = grid3d_soa(side=3, seed=2)
+    drv = DistributedRBCDDriver(ma, n, 2, Comm(), r=5,
+                                partition="contiguous", device="cpu")
+    res = drv.run(max_iters=150)
+    assert res.converged
+
+
+def test_soa_cpu_robust_rejected_cleanly():
+    from dpo_amd.comm import Comm
+    from dpo_amd.dist_driver import DistributedRBCDDriver
+    from dpo_amd.synthetic import grid3d_soa
+    ma, n = grid3d_soa(side=3, seed=2)
+    with pytest.raises(ValueError, match="cuda"):
+        DistributedRBCDDriver(ma, n, 2, Comm(), r=5,
+                              partition="contiguous", device="cpu",
+                              robust=RobustCostType.GNC_TLS)
