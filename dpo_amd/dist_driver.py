@@ -930,6 +930,27 @@ class DistributedRBCDDriver:
         self._sync_anchor()
         return res
 
+    def snapshot_initial_state(self) -> None:
+        """Record every local agent's current iterate so repeated
+        benchmark episodes can re-run the identical solve-to-target
+        problem (bench.py). Call once after construction."""
+        self._X0 = {rb: a.X.detach().clone()
+                    for rb, a in self.local_agents.items()}
+
+    def restore_initial_state(self) -> None:
+        """Reset iterates (and Nesterov state) to the snapshot; the next
+        run() starts from the same cold optimization state. Packed-path
+        buffers are updated in place, so cached hipGraphs stay valid."""
+        for rb, a in self.local_agents.items():
+            a.X.copy_(self._X0[rb])
+            if a.params.acceleration and a.Y is not None:
+                a.Y.copy_(a.X)
+                a.V.copy_(a.X)
+                a.gamma = 0.0
+                a.alpha = 0.0
+            a.iteration_number = 0
+        self._round_counter = 0
+
     def gather_final_trajectory(self):
         """Rounded global trajectory (d, (d+1) n) on rank 0 (reference
         PartitionInitial.cpp:329-335 output). Returns None on other
