@@ -409,3 +409,42 @@ def test_group_fanout_matches_per_agent_path(monkeypatch):
     for (ca, ga), (cb, gb) in zip(a.trace, b.trace):
         assert abs(ca - cb) < 1e-6 * max(1.0, abs(cb))
         assert abs(ga - gb) < 1e-4 * max(1.0, gb)
+
+
+def test_greedy_inner_tol0_stable():
+    """Regression (round-1 VERDICT weak #2): greedy selection with
+    inner_tol=0 must stay numerically stable — the monotone-acceptance
+    guard in the device solver must prevent divergence even when the
+    solver runs its full tCG budget every round on a converged state."""
+    from dpo_amd.comm import Comm
+    from dpo_amd.dist_driver import DistributedRBCDDriver
+    from dpo_amd.synthetic import grid3d
+    meas, n = grid3d(side=4, seed=0)
+    drv = DistributedRBCDDriver(meas, n, 4, Comm(), r=5,
+                                partition="contiguous",
+                                selection="greedy", device=DEV,
+                                inner_tol=0.0)
+    res = drv.run(max_iters=400, gradnorm_tol=0.0)
+    costs = [c for c, _ in res.trace]
+    assert np.isfinite(res.final_gradnorm)
+    assert res.final_gradnorm < 1.0, res.final_gradnorm
+    assert costs[-1] <= costs[0] + 1e-9
+
+
+def test_bench_episode_restore_deterministic():
+    """snapshot/restore (bench.py episodes) reproduces the identical
+    trajectory on the packed GPU path."""
+    from dpo_amd.comm import Comm
+    from dpo_amd.dist_driver import DistributedRBCDDriver
+    from dpo_amd.synthetic import grid3d
+    meas, n = grid3d(side=4, seed=1)
+    drv = DistributedRBCDDriver(meas, n, 4, Comm(), r=5,
+                                partition="contiguous",
+                                selection="colored", device=DEV)
+    drv.snapshot_initial_state()
+    r1 = drv.run(max_iters=1000, gradnorm_tol=0.1)
+    drv.restore_initial_state()
+    r2 = drv.run(max_iters=1000, gradnorm_tol=0.1)
+    assert r1.iterations == r2.iterations
+    assert r1.trace[0] == r2.trace[0]
+    assert r1.trace[-1] == r2.trace[-1]
