@@ -49,14 +49,23 @@ class TorchDistComm(Comm):
 
     def all_gather_flat(self, local: Tensor, sizes: List[int]) -> List[Tensor]:
         """All-gather variable-size fp64 flats (sizes known per rank).
-        Collectives need equal sizes -> pad to max (payloads are tiny)."""
+        Collectives need equal sizes -> pad to max (payloads are tiny).
+        Send/recv buffers are persistent (keyed by max size) so the
+        per-round path allocates nothing and RCCL sees stable pointers."""
         mx = max(sizes)
-        buf = torch.zeros(mx, dtype=local.dtype, device=self.device)
-        buf[:local.numel()] = local.to(self.device)
-        outs = [torch.empty(mx, dtype=local.dtype, device=self.device)
-                for _ in sizes]
-        dist.all_gather(outs, buf)
-        return [o[:s] for o, s in zip(outs, sizes)]
+        cache = getattr(self, "_ag_cache", None)
+        if cache is None or cache[0] != mx or cache[1].dtype != local.dtype:
+            buf = torch.zeros(mx, dtype=local.dtype, device=self.device)
+            out = torch.empty(self.world_size * mx, dtype=local.dtype,
+                              device=self.device)
+            self._ag_cache = cache = (mx, buf, out)
+        _, buf, out = cache
+        n = local.numel()
+        buf[:n] = local.to(self.device)
+        if n < mx:
+            buf[n:] = 0.0
+        dist.all_gather_into_tensor(out, buf)
+        return [out[i * mx:i * mx + s] for i, s in enumerate(sizes)]
 
     def all_reduce_sum_(self, t: Tensor) -> Tensor:
         tt = t.to(self.device)

@@ -446,5 +446,9 @@ def test_bench_episode_restore_deterministic():
     drv.restore_initial_state()
     r2 = drv.run(max_iters=1000, gradnorm_tol=0.1)
     assert r1.iterations == r2.iterations
-    assert r1.trace[0] == r2.trace[0]
-    assert r1.trace[-1] == r2.trace[-1]
+    # the first run's round-0 eval runs once eagerly (hipGraph capture
+    # warmup) while the restored run replays the graph — identical math,
+    # but reduction fan-in order can differ at the few-ULP level
+    for a, b in (r1.trace[0], r2.trace[0]), (r1.trace[-1], r2.trace[-1]):
+        assert abs(a[0] - b[0]) <= 1e-9 * max(1.0, abs(a[0]))
+        assert abs(a[1] - b[1]) <= 1e-9 * max(1.0, abs(a[1]))
