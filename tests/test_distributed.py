@@ -23,10 +23,12 @@ def _worker(rank, world, result_dir, mode):
 
     meas, n = grid3d(side=4, seed=0)
     comm = TorchDistComm("cpu")
+    sel = "colored" if mode == "accel_colored" else \
+        ("greedy" if mode == "accel" else mode)
     drv = DistributedRBCDDriver(
         meas, n, 4, comm, r=5, partition="contiguous",
-        selection=mode,
-        acceleration=(mode == "accel"))
+        selection=sel,
+        acceleration=mode.startswith("accel"))
     res = drv.run(max_iters=250)
     if rank == 0:
         with open(os.path.join(result_dir, "res.json"), "w") as f:
@@ -37,8 +39,12 @@ def _worker(rank, world, result_dir, mode):
 
 
 def _run_world2(mode):
+    return _run_world(2, mode)
+
+
+def _run_world(world, mode):
     with tempfile.TemporaryDirectory() as td:
-        mp.spawn(_worker, args=(2, td, mode), nprocs=2, join=True)
+        mp.spawn(_worker, args=(world, td, mode), nprocs=world, join=True)
         with open(os.path.join(td, "res.json")) as f:
             return json.load(f)
 
@@ -48,10 +54,12 @@ def _run_single(mode):
     from dpo_amd.dist_driver import DistributedRBCDDriver
     from dpo_amd.synthetic import grid3d
     meas, n = grid3d(side=4, seed=0)
+    sel = "colored" if mode == "accel_colored" else \
+        ("greedy" if mode == "accel" else mode)
     drv = DistributedRBCDDriver(
         meas, n, 4, Comm(), r=5, partition="contiguous",
-        selection=mode if mode != "accel" else "greedy",
-        acceleration=(mode == "accel"))
+        selection=sel,
+        acceleration=mode.startswith("accel"))
     return drv.run(max_iters=250)
 
 
@@ -69,3 +77,15 @@ def test_world2_matches_single_process(mode):
 def test_world2_accelerated():
     out = _run_world2("accel")
     assert out["conv"]
+
+
+def test_world4_accel_colored_combined_payload():
+    """World=4 (one agent per rank), accelerated + colored: exercises the
+    folded [X|Y] single-collective exchange layout on the dict path and
+    multi-rank agent spreading (8-GPU readiness)."""
+    out = _run_world(4, "accel_colored")
+    ref = _run_single("accel_colored")
+    assert out["conv"] == ref.converged
+    assert out["iters"] == ref.iterations
+    assert abs(out["cost"] - ref.final_cost) \
+        < 1e-6 * max(1, abs(ref.final_cost))
