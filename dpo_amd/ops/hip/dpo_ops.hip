@@ -541,6 +541,98 @@ __global__ void k_hess_fused(const int* __restrict__ row_ptr,
 }
 
 // ---------------------------------------------------------------------
+// Wide (element-per-thread) variant of k_hess_fused for LARGE agents.
+// One thread per pose-tile ELEMENT (dh*r threads per pose, 256/tile
+// poses per block): a 125k-pose agent then launches ~39k waves instead
+// of the ~2k a thread-per-pose kernel gets (which is < 2 waves/SIMD on
+// 256 CUs — too few to hide HBM latency, the round-1 1M-pose
+// bottleneck). The SpMM phase reads each neighbor block cooperatively;
+// the cross-element tangent projection stages the acc tile and X tile
+// through LDS. Same math and dot-product contract as k_hess_fused.
+// ---------------------------------------------------------------------
+template <int D, int R, int MODE, int CF = CF_NONE>
+__global__ void k_hess_wide(const int* __restrict__ row_ptr,
+                            const int* __restrict__ col_idx,
+                            const double* __restrict__ vals,
+                            const double* __restrict__ V,
+                            const double* __restrict__ X,
+                            const double* __restrict__ G,
+                            double* __restrict__ out,
+                            const double* __restrict__ dotW,
+                            double* __restrict__ ctrl,
+                            int n, int dot_slot, int dot_slot2,
+                            int guard) {
+  if (guarded_off(ctrl, guard)) return;
+  constexpr int dh = D + 1;
+  constexpr int TILE = dh * R;
+  constexpr int PB = 256 / TILE;  // poses per block
+  __shared__ double sAcc[PB][TILE];
+  __shared__ double sX[PB][TILE];
+  const int slot = threadIdx.x / TILE;
+  const int e = threadIdx.x % TILE;
+  const int c = e / R;            // row inside the dh x R tile
+  const int k = e % R;            // column
+  const bool active = threadIdx.x < PB * TILE;
+  const int i = blockIdx.x * PB + slot;
+  const bool live = active && i < n;
+  double acc = 0.0;
+  double d0 = 0.0, d1 = 0.0;
+  if (live) {
+    const int s0 = row_ptr[i], e0 = row_ptr[i + 1];
+    for (int p = s0; p < e0; ++p) {
+      const int j = col_idx[p];            // uniform across the tile
+      const double* B = vals + (size_t)p * dh * dh + c * dh;
+      const double* Vj = V + (size_t)j * dh * R + k;
+      #pragma unroll
+      for (int cc = 0; cc < dh; ++cc)
+        acc = fma(B[cc], Vj[cc * R], acc);
+    }
+    const double g = G ? G[(size_t)i * TILE + e] : 0.0;
+    acc += g;
+    if (MODE == 1) {
+      const double v = V[(size_t)i * TILE + e];
+      d0 = (acc - g) * v;
+      d1 = g * v;
+    } else {
+      const double x = X[(size_t)i * TILE + e];
+      if (dot_slot2 >= 0) d1 = acc * x;
+      sAcc[slot][e] = acc;
+      sX[slot][e] = x;
+    }
+  }
+  if (MODE == 0) {
+    __syncthreads();
+    double o = acc;
+    if (live && c < D) {
+      #pragma unroll
+      for (int b = 0; b < D; ++b) {
+        double s1 = 0.0, s2 = 0.0;
+        #pragma unroll
+        for (int kk = 0; kk < R; ++kk) {
+          s1 = fma(sX[slot][c * R + kk], sAcc[slot][b * R + kk], s1);
+          s2 = fma(sX[slot][b * R + kk], sAcc[slot][c * R + kk], s2);
+        }
+        o = fma(-0.5 * (s1 + s2), sX[slot][b * R + k], o);
+      }
+    }
+    if (live) {
+      out[(size_t)i * TILE + e] = o;
+      if (dot_slot >= 0) {
+        const double w = dotW ? dotW[(size_t)i * TILE + e] : o;
+        d0 = o * w;
+      }
+    }
+    __syncthreads();  // sAcc/sX reuse barrier for CF tail safety
+  }
+  if (dot_slot >= 0) block_reduce_atomic(d0, ctrl + dot_slot);
+  if (dot_slot2 >= 0) block_reduce_atomic(d1, ctrl + dot_slot2);
+  if (CF != CF_NONE) {
+    if (fanin_last_block(ctrl) && threadIdx.x == 0)
+      run_ctrl_tail(CF, ctrl, nullptr);
+  }
+}
+
+// ---------------------------------------------------------------------
 // Batched polar projection onto (St(d, r) x R^r)^n of an affine
 // combination  M = ca*A + cb*B + cc*C  (B, C optional).
 // polar(Mt) for the wide d x r Stiefel block via the analytic
@@ -1894,6 +1986,21 @@ static void launch_polar(const double* A, const double* B, const double* C,
   dpo_bad_shape(d, r);
 }
 
+// Wide-variant selection: element-per-thread (dh*r threads per pose)
+// for large agents, where the thread-per-pose kernel's n/64 waves are
+// too few to hide HBM latency (1M-pose profile, round 1); the fused
+// thread-per-pose kernel stays the default for the (launch-latency-
+// bound) small-agent regime. DPO_HESS_WIDE=0/1 forces either path.
+static inline bool hess_use_wide(int n) {
+  static const int mode = []() {
+    const char* v = getenv("DPO_HESS_WIDE");
+    return v ? atoi(v) : -1;
+  }();
+  if (mode == 0) return false;
+  if (mode == 1) return true;
+  return n >= 16384;
+}
+
 template <int MODE, int CF = CF_NONE>
 static void launch_hess_fused(const int* rp, const int* ci,
                               const double* vals, const double* V,
@@ -1903,11 +2010,20 @@ static void launch_hess_fused(const int* rp, const int* ci,
                               int dot_slot, int dot_slot2, int guard,
                               hipStream_t s) {
   const int grid = blocks_for(n, 256);
+  const bool wide = hess_use_wide(n);
 #define CASE_HF(D, R) \
   if (d == D && r == R) { \
-    hipLaunchKernelGGL((k_hess_fused<D, R, MODE, CF>), dim3(grid), \
-                       dim3(256), 0, s, rp, ci, vals, V, X, G, out, dotW, \
-                       ctrl, n, dot_slot, dot_slot2, guard); \
+    if (wide) { \
+      constexpr int PB = 256 / ((D + 1) * R); \
+      hipLaunchKernelGGL((k_hess_wide<D, R, MODE, CF>), \
+                         dim3(blocks_for(n, PB)), dim3(256), 0, s, \
+                         rp, ci, vals, V, X, G, out, dotW, \
+                         ctrl, n, dot_slot, dot_slot2, guard); \
+    } else { \
+      hipLaunchKernelGGL((k_hess_fused<D, R, MODE, CF>), dim3(grid), \
+                         dim3(256), 0, s, rp, ci, vals, V, X, G, out, \
+                         dotW, ctrl, n, dot_slot, dot_slot2, guard); \
+    } \
     return; \
   }
   DPO_FOREACH_DR(CASE_HF)

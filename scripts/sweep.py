@@ -16,6 +16,8 @@ def main():
     ap.add_argument("--device", default="cuda:0")
     ap.add_argument("--partition", default="contiguous")
     ap.add_argument("--robots", type=int, default=5)
+    ap.add_argument("--selection", default="greedy")
+    ap.add_argument("--accel", action="store_true")
     ap.add_argument("--max-iters", type=int, default=1000)
     ap.add_argument("--datasets", default="",
                     help="comma-separated subset (default: all)")
@@ -36,11 +38,13 @@ def main():
                 continue
             drv = DistributedRBCDDriver(
                 meas, n, args.robots, Comm(), r=5,
-                partition=args.partition, device=args.device)
+                partition=args.partition, device=args.device,
+                selection=args.selection, acceleration=args.accel)
             res = drv.run(max_iters=args.max_iters)
             print(json.dumps({
                 "dataset": name, "poses": n, "edges": len(meas),
-                "partition": args.partition,
+                "partition": args.partition, "selection": args.selection,
+                "accel": args.accel, "robots": args.robots,
                 "iterations": res.iterations, "converged": res.converged,
                 "final_cost": res.final_cost,
                 "final_gradnorm": res.final_gradnorm,

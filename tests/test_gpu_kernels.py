@@ -452,3 +452,38 @@ def test_bench_episode_restore_deterministic():
     for a, b in (r1.trace[0], r2.trace[0]), (r1.trace[-1], r2.trace[-1]):
         assert abs(a[0] - b[0]) <= 1e-9 * max(1.0, abs(a[0]))
         assert abs(a[1] - b[1]) <= 1e-9 * max(1.0, abs(a[1]))
+
+
+def test_hess_wide_matches_narrow():
+    """k_hess_wide (element-per-thread, large-agent layout) must produce
+    the same solve trajectory as k_hess_fused (thread-per-pose). Run in
+    subprocesses because the path choice is cached from the env."""
+    import json
+    import os
+    import subprocess
+    import sys
+    here = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    code = (
+        "import json\n"
+        "from dpo_amd.comm import Comm\n"
+        "from dpo_amd.dist_driver import DistributedRBCDDriver\n"
+        "from dpo_amd.synthetic import grid3d\n"
+        "meas, n = grid3d(side=6, seed=3)\n"
+        "drv = DistributedRBCDDriver(meas, n, 2, Comm(), r=5,\n"
+        "    partition='contiguous', selection='colored', device='cuda:0')\n"
+        "res = drv.run(max_iters=40, gradnorm_tol=0.0)\n"
+        "print(json.dumps({'cost': res.final_cost,\n"
+        "                  'gn': res.final_gradnorm,\n"
+        "                  't0': res.trace[0]}))\n")
+    outs = {}
+    for w in ("0", "1"):
+        env = dict(os.environ, DPO_HESS_WIDE=w)
+        r = subprocess.run([sys.executable, "-c", code],
+                           capture_output=True, text=True, env=env,
+                           cwd=here, timeout=600)
+        assert r.returncode == 0, r.stderr[-2000:]
+        outs[w] = json.loads(r.stdout.strip().splitlines()[-1])
+    a, b = outs["0"], outs["1"]
+    assert abs(a["t0"][0] - b["t0"][0]) < 1e-9 * max(1, abs(a["t0"][0]))
+    assert abs(a["cost"] - b["cost"]) < 1e-8 * max(1, abs(a["cost"]))
+    assert abs(a["gn"] - b["gn"]) < 1e-5 * max(1.0, abs(a["gn"]))
