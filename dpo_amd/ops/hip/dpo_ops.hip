@@ -633,6 +633,60 @@ __global__ void k_hess_wide(const int* __restrict__ row_ptr,
 }
 
 // ---------------------------------------------------------------------
+// fp64-MFMA BSR SpMM (A/B study, round-1 VERDICT item 4; d = 3 only).
+// One wave computes 4 pose rows as a 16x16 v_mfma_f64_16x16x4_f64 tile
+// accumulating over the 4-row group's COLUMN UNION (grouped-ELL built
+// host-side): A = the 4 poses' stacked 4x4 Q blocks for union column u
+// (zeros where a pose lacks that column), B = the neighbor pose block
+// V_j (cols 5..15 zero-padded). Useful-FLOP ratio is r/16 x the union
+// fill (~20-30% at r=5), and gfx950's fp64 matrix rate equals the
+// vector rate, so this is expected to LOSE to the scalar-FMA kernels —
+// the point is to measure that, with rocprof counters, instead of
+// arguing it.
+// ---------------------------------------------------------------------
+typedef double dpo_d4 __attribute__((ext_vector_type(4)));
+
+__global__ void k_bsr_spmm_mfma_d3(const int* __restrict__ grp_ptr,
+                                   const int* __restrict__ grp_cols,
+                                   const int* __restrict__ grp_blk,
+                                   const double* __restrict__ vals,
+                                   const double* __restrict__ X,
+                                   double* __restrict__ out,
+                                   int ngroups, int n, int r) {
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int g = blockIdx.x * (blockDim.x >> 6) + wave;
+  if (g >= ngroups) return;
+  const int row16 = lane & 15;        // A row: pose-in-group*4 + brow
+  const int kfrag = lane >> 4;        // K index 0..3
+  const int pose_in_g = row16 >> 2;
+  const int brow = row16 & 3;
+  dpo_d4 acc = {0.0, 0.0, 0.0, 0.0};
+  const int s0 = grp_ptr[g], e0 = grp_ptr[g + 1];
+  for (int u = s0; u < e0; ++u) {
+    const int j = grp_cols[u];
+    const int blk = grp_blk[(size_t)u * 4 + pose_in_g];
+    const double a = (blk >= 0)
+        ? vals[(size_t)blk * 16 + brow * 4 + kfrag] : 0.0;
+    const int col = lane & 15;
+    const double b = (col < r)
+        ? X[(size_t)j * 4 * r + kfrag * r + col] : 0.0;
+    acc = __builtin_amdgcn_mfma_f64_16x16x4f64(a, b, acc, 0, 0, 0);
+  }
+  // D layout: lane l, reg q -> D[row = 4*(l>>4)+q][col = l&15]
+  const int col = lane & 15;
+  if (col < r) {
+    #pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      const int row = 4 * (lane >> 4) + q;
+      const int pose = g * 4 + (row >> 2);
+      if (pose < n)
+        out[(size_t)pose * 4 * r + (row & 3) * r + col] = acc[q];
+    }
+  }
+}
+
+// ---------------------------------------------------------------------
 // Batched polar projection onto (St(d, r) x R^r)^n of an affine
 // combination  M = ca*A + cb*B + cc*C  (B, C optional).
 // polar(Mt) for the wide d x r Stiefel block via the analytic
@@ -2093,6 +2147,18 @@ void dpo_bsr_spmm(const int* row_ptr, const int* col_idx, const double* vals,
                   const double* ctrl, int guard, void* stream) {
   launch_spmm(row_ptr, col_idx, vals, n, dh - 1, r, X, out, ctrl, guard,
               (hipStream_t)stream);
+}
+
+void dpo_bsr_spmm_mfma(const int* grp_ptr, const int* grp_cols,
+                       const int* grp_blk, const double* vals,
+                       const double* X, double* out, int ngroups, int n,
+                       int r, void* stream) {
+  const int waves_per_block = 4;
+  const int grid = (ngroups + waves_per_block - 1) / waves_per_block;
+  hipLaunchKernelGGL(k_bsr_spmm_mfma_d3, dim3(grid),
+                     dim3(64 * waves_per_block), 0, (hipStream_t)stream,
+                     grp_ptr, grp_cols, grp_blk, vals, X, out, ngroups,
+                     n, r);
 }
 
 void dpo_proj_dots(const double* X, const double* V, const double* G,

@@ -28,6 +28,7 @@ _l = ctypes.c_long
 _d = ctypes.c_double
 
 _lib.dpo_bsr_spmm.argtypes = [_c, _c, _c, _i, _i, _c, _c, _i, _c, _i, _c]
+_lib.dpo_bsr_spmm_mfma.argtypes = [_c, _c, _c, _c, _c, _c, _i, _i, _i, _c]
 _lib.dpo_proj_dots.argtypes = [_c, _c, _c, _c, _c, _c, _i, _i, _i, _i, _i,
                                _i, _i, _c]
 _lib.dpo_polar_affine.argtypes = [_c, _c, _c, _d, _d, _d, _c, _i, _i, _i,
@@ -158,6 +159,49 @@ def bsr_spmm(row_ptr: Tensor, col_idx: Tensor, vals: Tensor, n: int,
         out = torch.empty_like(X)
     _lib.dpo_bsr_spmm(_p(row_ptr), _p(col_idx), _p(vals), n, dh,
                       _p(X), _p(out), r, _p(ctrl), guard, _stream(X))
+    return out
+
+
+def build_spmm_mfma_groups(row_ptr, col_idx, n: int):
+    """Host-side grouped-ELL structure for the fp64-MFMA SpMM A/B
+    kernel (d = 3): 4-pose row groups with per-group column unions.
+    Returns (grp_ptr, grp_cols, grp_blk) numpy int32 arrays; grp_blk has
+    4 entries per (group, union-column) = the Q block index of each pose
+    in the group for that column, or -1."""
+    import numpy as np
+    rp = row_ptr.cpu().numpy() if isinstance(row_ptr, Tensor) else row_ptr
+    ci = col_idx.cpu().numpy() if isinstance(col_idx, Tensor) else col_idx
+    ngroups = (n + 3) // 4
+    grp_ptr = [0]
+    grp_cols = []
+    grp_blk = []
+    for g in range(ngroups):
+        rows = range(g * 4, min(g * 4 + 4, n))
+        union = {}
+        for pi, i in enumerate(rows):
+            for p in range(rp[i], rp[i + 1]):
+                union.setdefault(int(ci[p]), [-1, -1, -1, -1])[pi] = p
+        for j in sorted(union):
+            grp_cols.append(j)
+            grp_blk.extend(union[j])
+        grp_ptr.append(len(grp_cols))
+    return (np.asarray(grp_ptr, dtype=np.int32),
+            np.asarray(grp_cols, dtype=np.int32),
+            np.asarray(grp_blk, dtype=np.int32))
+
+
+def bsr_spmm_mfma(grp_ptr: Tensor, grp_cols: Tensor, grp_blk: Tensor,
+                  vals: Tensor, X: Tensor,
+                  out: Optional[Tensor] = None) -> Tensor:
+    """fp64-MFMA SpMM (d=3 / dh=4 only); A/B study kernel."""
+    _chk(vals); _chk(X)
+    r = X.shape[1]
+    n = X.shape[0] // 4
+    if out is None:
+        out = torch.empty_like(X)
+    _lib.dpo_bsr_spmm_mfma(_p(grp_ptr), _p(grp_cols), _p(grp_blk),
+                           _p(vals), _p(X), _p(out),
+                           int(grp_ptr.numel() - 1), n, r, _stream(X))
     return out
 
 
