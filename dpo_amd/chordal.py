@@ -212,3 +212,157 @@ def odometry_initialization(d: int, num_poses: int,
             T[:, dst * dh:dst * dh + d] = Rsrc @ m.R
             T[:, dst * dh + d] = tsrc + Rsrc @ m.t
     return T
+
+
+# ---------------------------------------------------------------------
+# SoA / GPU-native chordal initialization (no B matrices, no Python
+# per-edge loops). Mathematically equivalent to the reference's SPQR
+# least-squares (DPGO_utils.cpp:273-409): the rotation stage solves the
+# normal equations of min sum_e kappa_e ||R_j - R_i R_e||_F^2 with pose
+# 0 pinned — i.e. the rotation connection Laplacian system — by
+# preconditioned CG whose SpMV is our block-CSR HIP kernel; rotations
+# are rounded to SO(d) by batched SVD; the translation stage solves the
+# tau-weighted scalar graph Laplacian the same way. Works on the
+# MeasurementArray directly, so a 1M-pose chordal init assembles in
+# vectorized numpy and iterates on the GPU.
+# ---------------------------------------------------------------------
+def _bsr_from_coo(rows, cols, blocks, n, dh, device="cpu"):
+    import torch
+    from .quadratic import BSRMatrix
+    order = np.lexsort((cols, rows))
+    rows, cols = rows[order], cols[order]
+    blocks = blocks[order]
+    key = rows * n + cols
+    uniq, inv = np.unique(key, return_inverse=True)
+    acc = np.zeros((len(uniq), dh, dh))
+    np.add.at(acc, inv, blocks)
+    urows = (uniq // n).astype(np.int64)
+    ucols = (uniq % n).astype(np.int32)
+    row_ptr = np.zeros(n + 1, dtype=np.int32)
+    np.add.at(row_ptr, urows + 1, 1)
+    row_ptr = np.cumsum(row_ptr, dtype=np.int64).astype(np.int32)
+    dev = torch.device(device)
+    return BSRMatrix(n, dh,
+                     torch.from_numpy(row_ptr).to(dev),
+                     torch.from_numpy(ucols).to(dev),
+                     torch.from_numpy(acc).to(dev))
+
+
+def _pcg_block(A, rhs, diag_inv, tol, max_iters):
+    """CG on A x = rhs (x: (N, r) torch), Jacobi-scaled."""
+    import torch
+    x = torch.zeros_like(rhs)
+    r = rhs.clone()
+    z = r * diag_inv
+    p = z.clone()
+    rz = float((r * z).sum())
+    rhs_n = float((rhs * rhs).sum())
+    if rhs_n == 0.0:
+        return x, 0
+    it = 0
+    for it in range(max_iters):
+        Ap = A.spmm(p)
+        pAp = float((p * Ap).sum())
+        if pAp <= 0:
+            break
+        alpha = rz / pAp
+        x += alpha * p
+        r -= alpha * Ap
+        if float((r * r).sum()) <= tol * tol * rhs_n:
+            break
+        z = r * diag_inv
+        rz_new = float((r * z).sum())
+        p = z + (rz_new / rz) * p
+        rz = rz_new
+    return x, it + 1
+
+
+def chordal_initialization_soa(ma, num_poses: int, device: str = "cpu",
+                               tol: float = 1e-8,
+                               max_iters: int = 1000) -> np.ndarray:
+    """Chordal initialization from a MeasurementArray (global indices).
+    Returns T (d, (d+1) n) with pose 0 = identity, like
+    chordal_initialization."""
+    import torch
+    d = ma.d
+    n = num_poses
+    p1 = ma.p1.astype(np.int64)
+    p2 = ma.p2.astype(np.int64)
+    kap = ma.kappa
+    Re = ma.R  # (ne, d, d)
+    Id = np.eye(d)
+
+    # ---- rotation stage: unknowns X_i = R_i^T, poses 1..n-1 ----------
+    # edge (i,j): A_ii += k I, A_jj += k I, A_ij -= k R_e, A_ji -= k R_e^T
+    m = n - 1  # unknown poses (shifted by -1)
+    i_u = p1 - 1
+    j_u = p2 - 1
+    rows_l, cols_l, blks_l = [], [], []
+
+    both = (p1 > 0) & (p2 > 0)
+    for sel, rr, cc, bb in (
+            (p1 > 0, i_u, i_u, kap[:, None, None] * Id[None]),
+            (p2 > 0, j_u, j_u, kap[:, None, None] * Id[None]),
+            (both, i_u, j_u, -kap[:, None, None] * Re),
+            (both, j_u, i_u,
+             -kap[:, None, None] * np.transpose(Re, (0, 2, 1)))):
+        rows_l.append(rr[sel])
+        cols_l.append(cc[sel])
+        blks_l.append(np.ascontiguousarray(bb[sel]))
+    A = _bsr_from_coo(np.concatenate(rows_l), np.concatenate(cols_l),
+                      np.concatenate(blks_l), m, d, device)
+    rhs = np.zeros((m, d, d))
+    s = (p1 == 0)
+    if s.any():  # edge (0, j): RHS_j += k R_e^T
+        np.add.at(rhs, j_u[s],
+                  kap[s, None, None] * np.transpose(Re[s], (0, 2, 1)))
+    s = (p2 == 0)
+    if s.any():  # edge (i, 0): RHS_i += k R_e
+        np.add.at(rhs, i_u[s], kap[s, None, None] * Re[s])
+    dev = torch.device(device)
+    rhs_t = torch.from_numpy(rhs.reshape(m * d, d)).to(dev)
+    dinv = 1.0 / A.diag_blocks()[:, 0, 0]  # diagonal blocks are c*I
+    dinv_t = dinv.repeat_interleave(d).unsqueeze(1)
+    X, _ = _pcg_block(A, rhs_t, dinv_t, tol, max_iters)
+    # round to SO(d): R_i = proj(X_i^T), batched SVD on the device
+    Xt = X.view(m, d, d).transpose(1, 2).contiguous()
+    U, _, Vh = torch.linalg.svd(Xt)
+    det = torch.linalg.det(U @ Vh)
+    D = torch.ones(m, d, dtype=torch.float64, device=dev)
+    D[:, -1] = torch.sign(det + (det == 0))
+    Rall = (U * D.unsqueeze(1)) @ Vh  # (m, d, d)
+    R_full = torch.empty(n, d, d, dtype=torch.float64, device=dev)
+    R_full[0] = torch.eye(d, dtype=torch.float64, device=dev)
+    R_full[1:] = Rall
+
+    # ---- translation stage: scalar tau-Laplacian, d RHS columns ------
+    tau = ma.tau
+    c_e = torch.einsum(
+        "eij,ej->ei", R_full.index_select(0, torch.from_numpy(p1).to(dev)),
+        torch.from_numpy(ma.t).to(dev)).cpu().numpy()  # R_i t_e
+    rows_l, cols_l, vals_l = [], [], []
+    for sel, rr, cc, vv in (
+            (p1 > 0, i_u, i_u, tau),
+            (p2 > 0, j_u, j_u, tau),
+            (both, i_u, j_u, -tau),
+            (both, j_u, i_u, -tau)):
+        rows_l.append(rr[sel])
+        cols_l.append(cc[sel])
+        vals_l.append(vv[sel].reshape(-1, 1, 1))
+    At = _bsr_from_coo(np.concatenate(rows_l), np.concatenate(cols_l),
+                       np.concatenate(vals_l), m, 1, device)
+    rhs_tr = np.zeros((m, d))
+    s1 = (p1 > 0)
+    np.add.at(rhs_tr, i_u[s1], -(tau[s1, None] * c_e[s1]))
+    s2 = (p2 > 0)
+    np.add.at(rhs_tr, j_u[s2], tau[s2, None] * c_e[s2])
+    rhs_tr_t = torch.from_numpy(rhs_tr).to(dev)
+    dinv_tr = (1.0 / At.diag_blocks()[:, 0, 0]).unsqueeze(1)
+    tsol, _ = _pcg_block(At, rhs_tr_t, dinv_tr, tol, max_iters)
+
+    dh = d + 1
+    T = np.zeros((d, n * dh))
+    Tv = T.reshape(d, n, dh).transpose(1, 0, 2)  # (n, d, dh) view
+    Tv[:, :, :d] = R_full.cpu().numpy()
+    Tv[1:, :, d] = tsol.cpu().numpy()
+    return T

@@ -163,12 +163,20 @@ class DistributedRBCDDriver:
         from .manifold import lifting_matrix
         YL = lifting_matrix(d, r)
         if self._soa:
-            # global-frame initialization: chordal is too heavy at SoA
-            # scale, so distribute an externally provided warm start
-            # (ma.warm_start) or the GLOBAL odometry dead-reckoning
-            # (prefix scan, vectorized) — consistent frames across agents.
+            # global-frame initialization: an externally provided warm
+            # start (ma.warm_start) wins; else L2 mode runs the SoA/GPU
+            # chordal init (vectorized assembly + BSR-kernel PCG —
+            # reference MultiRobotExample.cpp:185-202 semantics at SoA
+            # scale); robust mode keeps the odometry dead-reckoning
+            # prefix scan (reference PGOAgent.cpp:952-957 does not trust
+            # loop closures before GNC).
             if getattr(measurements, "warm_start", None) is not None:
                 T_chordal_pre = measurements.warm_start
+            elif robust == RobustCostType.L2:
+                from .chordal import chordal_initialization_soa
+                T_chordal_pre = chordal_initialization_soa(
+                    measurements, num_poses, device=device,
+                    tol=1e-6, max_iters=500)
             else:
                 from .measurements import odometry_initialization_array
                 odo_mask = (measurements.p1 + 1 == measurements.p2)
