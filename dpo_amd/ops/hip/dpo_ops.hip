@@ -415,6 +415,72 @@ __global__ void k_proj_dots(const double* __restrict__ X,
   }
 }
 
+// Wide (element-per-thread) tangent projection for LARGE agents: same
+// wave-starvation fix as k_hess_wide (a thread-per-pose kernel gets
+// only n/64 waves). dh*r threads per pose; the cross-element sym(Y V^T)
+// contraction goes through LDS tiles.
+template <int NEG, int D, int R, int CF = CF_NONE>
+__global__ void k_proj_wide(const double* __restrict__ X,
+                            const double* __restrict__ V,
+                            const double* __restrict__ G,
+                            double* __restrict__ out,
+                            const double* __restrict__ dotWith,
+                            double* __restrict__ ctrl,
+                            int n, int dot_slot, int dot_slot2,
+                            int guard) {
+  if (guarded_off(ctrl, guard)) return;
+  constexpr int dh = D + 1;
+  constexpr int TILE = dh * R;
+  constexpr int PB = 256 / TILE;
+  __shared__ double sV[PB][TILE];
+  __shared__ double sX[PB][TILE];
+  const int slot = threadIdx.x / TILE;
+  const int e = threadIdx.x % TILE;
+  const int c = e / R;
+  const int k = e % R;
+  const bool active = threadIdx.x < PB * TILE;
+  const int i = blockIdx.x * PB + slot;
+  const bool live = active && i < n;
+  double v = 0.0, x = 0.0, d0 = 0.0, d1 = 0.0;
+  if (live) {
+    v = V[(size_t)i * TILE + e];
+    if (G) v += G[(size_t)i * TILE + e];
+    x = X[(size_t)i * TILE + e];
+    if (dot_slot2 >= 0) d1 = v * x;
+    sV[slot][e] = v;
+    sX[slot][e] = x;
+  }
+  __syncthreads();
+  double o = v;
+  if (live && c < D) {
+    #pragma unroll
+    for (int b = 0; b < D; ++b) {
+      double s1 = 0.0, s2 = 0.0;
+      #pragma unroll
+      for (int kk = 0; kk < R; ++kk) {
+        s1 = fma(sX[slot][c * R + kk], sV[slot][b * R + kk], s1);
+        s2 = fma(sX[slot][b * R + kk], sV[slot][c * R + kk], s2);
+      }
+      o = fma(-0.5 * (s1 + s2), sX[slot][b * R + k], o);
+    }
+  }
+  if (live) {
+    const double p = NEG ? -o : o;
+    out[(size_t)i * TILE + e] = p;
+    if (dot_slot >= 0) {
+      const double w = dotWith ? dotWith[(size_t)i * TILE + e] : o;
+      d0 = p * w;
+    }
+  }
+  __syncthreads();
+  if (dot_slot >= 0) block_reduce_atomic(d0, ctrl + dot_slot);
+  if (dot_slot2 >= 0) block_reduce_atomic(d1, ctrl + dot_slot2);
+  if (CF != CF_NONE) {
+    if (fanin_last_block(ctrl) && threadIdx.x == 0)
+      run_ctrl_tail(CF, ctrl, nullptr);
+  }
+}
+
 // ---------------------------------------------------------------------
 // Fused Hessian-vector product: one THREAD per pose computes its whole
 // BSR row of Q @ V (the dh x r accumulator tile lives in registers; each
@@ -1997,17 +2063,28 @@ static void launch_spmm(const int* rp, const int* ci, const double* vals,
 }
 
 template <int CF>
+static inline bool hess_use_wide(int n);
+
 static void launch_proj_dots_cf(const double* X, const double* V,
                                 const double* G, double* out,
                                 const double* dotWith, double* ctrl,
                                 int n, int d, int r, int dot_slot,
                                 int dot_slot2, int guard, hipStream_t s) {
   const int grid = blocks_for(n, 256);
+  const bool wide = hess_use_wide(n);
 #define CASE_PROJ(D, R) \
   if (d == D && r == R) { \
-    hipLaunchKernelGGL((k_proj_dots<0, D, R, CF>), dim3(grid), dim3(256), \
-                       0, s, X, V, G, out, dotWith, ctrl, n, dot_slot, \
-                       dot_slot2, guard); \
+    if (wide) { \
+      constexpr int PB = 256 / ((D + 1) * R); \
+      hipLaunchKernelGGL((k_proj_wide<0, D, R, CF>), \
+                         dim3(blocks_for(n, PB)), dim3(256), 0, s, \
+                         X, V, G, out, dotWith, ctrl, n, dot_slot, \
+                         dot_slot2, guard); \
+    } else { \
+      hipLaunchKernelGGL((k_proj_dots<0, D, R, CF>), dim3(grid), \
+                         dim3(256), 0, s, X, V, G, out, dotWith, ctrl, \
+                         n, dot_slot, dot_slot2, guard); \
+    } \
     return; \
   }
   DPO_FOREACH_DR(CASE_PROJ)
