@@ -457,6 +457,13 @@ class DistributedRBCDDriver:
                 color = it % self._num_colors
                 active = [rb for rb in range(self.num_robots)
                           if self._colors[rb] == color]
+            elif self.selection == "colored_greedy":
+                # activate the whole independent set (color) that
+                # contains the max-gradient agent: greedy's targeting
+                # with colored's concurrent block updates
+                color = self._colors[selected]
+                active = [rb for rb in range(self.num_robots)
+                          if self._colors[rb] == color]
             elif self.selection == "round_robin":
                 active = [it % self.num_robots]
             else:
@@ -637,7 +644,8 @@ class DistributedRBCDDriver:
         if modes is None:
             # env decisions are fixed per driver (read once, not per round)
             import os as _os
-            _sync_eval_default = "0" if self.selection == "colored" else "1"
+            _sync_eval_default = ("0" if self.selection in (
+                "colored", "colored_greedy") else "1")
             modes = self._eval_env_modes = (
                 _os.environ.get("DPO_DRIVER_EVAL_GRAPH", "0") == "1",
                 _os.environ.get("DPO_SYNC_EVAL", _sync_eval_default) == "1")
@@ -785,7 +793,7 @@ class DistributedRBCDDriver:
                            and not _sync_solve)
         inner_tol = next(iter(self.local_agents.values())) \
             .params.inner_tol if self.local_agents else 1e-2
-        if self.selection == "colored":
+        if self.selection in ("colored", "colored_greedy"):
             color_active = [
                 [rb for rb in range(self.num_robots)
                  if self._colors[rb] == c]
@@ -796,7 +804,8 @@ class DistributedRBCDDriver:
                                if rb in self.local_agents])
                     for ca in color_active]
         chunk = 16 if (gradnorm_tol <= 0.0
-                       and self.selection != "greedy"
+                       and self.selection not in ("greedy",
+                                                  "colored_greedy")
                        and _os_.environ.get("DPO_DRIVER_EVAL_GRAPH",
                                             "0") != "1") else 1
         evalring = (torch.zeros(chunk, self.num_robots, 3,
@@ -848,6 +857,8 @@ class DistributedRBCDDriver:
                     a.robust_cost.update()
             if self.selection == "colored":
                 active = color_active[it % self._num_colors]
+            elif self.selection == "colored_greedy":
+                active = color_active[self._colors[selected]]
             elif self.selection == "round_robin":
                 selected = it % self.num_robots
                 active = [selected]
@@ -868,6 +879,8 @@ class DistributedRBCDDriver:
             if use_group_solve:
                 if self.selection == "colored":
                     gids = color_ids[it % self._num_colors]
+                elif self.selection == "colored_greedy":
+                    gids = color_ids[self._colors[selected]]
                 elif selected in self.local_agents:
                     gids = group.ids([self._group_lidx[selected]])
                 else:

@@ -2062,9 +2062,9 @@ static void launch_spmm(const int* rp, const int* ci, const double* vals,
                      rp, ci, vals, X, out, n, dh, r, ctrl, guard);
 }
 
-template <int CF>
 static inline bool hess_use_wide(int n);
 
+template <int CF>
 static void launch_proj_dots_cf(const double* X, const double* V,
                                 const double* G, double* out,
                                 const double* dotWith, double* ctrl,
@@ -2725,10 +2725,12 @@ static void enqueue_solve_body(DpoCtx* c, double* X, const double* nbr,
   if (nbr) ctx_assemble_g(c, nbr, s);
   dzero(c->ctrl, CTRL_SIZE, s);
 
-  // gradient phase
-  ctx_spmm(c, X, c->W, -1, s);
-  launch_proj_dots(X, c->W, c->Gt, c->grad, nullptr, c->ctrl, n,
-                   d, r, C_DOT1, C_DOT0, -1, s);
+  // gradient phase: one fused kernel computes Q@X + G, projects at X
+  // and accumulates <P,P> / <QX+G,X> (was spmm + proj_dots — the
+  // two-kernel split re-read the full iterate from HBM)
+  launch_hess_fused<0>(c->q_rp, c->q_ci, c->q_vals, X, X, c->Gt,
+                       c->grad, nullptr, c->ctrl, n, d, r,
+                       C_DOT1, C_DOT0, -1, s);
   if (c->Gt)
     hipLaunchKernelGGL(k_dots, dim3(gvec), dim3(256), 0, s,
                        c->Gt, X, (const double*)nullptr, c->ctrl,
@@ -3039,9 +3041,10 @@ void dpo_eval_terms(void* h, const double* X, double* out_dev,
   const int gvec = (int)((total + 255) / 256);
   static const bool no_cf = dpo_env_flag("DPO_NO_CF");
   dzero(c->ctrl + C_DOT0, 4, s);
-  ctx_spmm(c, X, c->W, -1, s);
-  launch_proj_dots(X, c->W, c->Gt, c->grad, nullptr, c->ctrl, n,
-                   d, r, C_DOT1, C_DOT0, -1, s);
+  // fused: Q@X + G, projection at X, <P,P> and <QX+G,X> in one kernel
+  launch_hess_fused<0>(c->q_rp, c->q_ci, c->q_vals, X, X, c->Gt,
+                       c->grad, nullptr, c->ctrl, n, d, r,
+                       C_DOT1, C_DOT0, -1, s);
   if (c->Gt && !no_cf) {
     // the <G, X> dot's last block combines all three scalars into
     // out_dev (CF_COMBINE tail) instead of a separate kernel
