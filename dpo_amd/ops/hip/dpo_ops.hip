@@ -502,12 +502,12 @@ __global__ void k_hess_fused(const int* __restrict__ row_ptr,
                              const double* __restrict__ dotW,
                              double* __restrict__ ctrl,
                              int n, int dot_slot, int dot_slot2,
-                             int guard) {
+                             int dot_slot3, int guard) {
   if (guarded_off(ctrl, guard)) return;
   constexpr int dh = D + 1;
   const int i = blockIdx.x * blockDim.x + threadIdx.x;
   double acc[dh][R];
-  double d0 = 0.0, d1 = 0.0;
+  double d0 = 0.0, d1 = 0.0, d2 = 0.0;
   if (i < n) {
     #pragma unroll
     for (int c = 0; c < dh; ++c)
@@ -529,10 +529,16 @@ __global__ void k_hess_fused(const int* __restrict__ row_ptr,
     }
     if (G) {
       const double* Gi = G + (size_t)i * dh * R;
+      const double* Xi3 = (MODE == 0 && dot_slot3 >= 0)
+          ? X + (size_t)i * dh * R : nullptr;
       #pragma unroll
       for (int c = 0; c < dh; ++c)
         #pragma unroll
-        for (int k = 0; k < R; ++k) acc[c][k] += Gi[c * R + k];
+        for (int k = 0; k < R; ++k) {
+          const double g = Gi[c * R + k];
+          acc[c][k] += g;
+          if (Xi3) d2 = fma(g, Xi3[c * R + k], d2);
+        }
     }
     if (MODE == 1) {
       // dots only: d0 = <(QV)_i + G_i, V_i>, d1 = <G_i, V_i>
@@ -600,6 +606,8 @@ __global__ void k_hess_fused(const int* __restrict__ row_ptr,
   }
   if (dot_slot >= 0) block_reduce_atomic(d0, ctrl + dot_slot);
   if (dot_slot2 >= 0) block_reduce_atomic(d1, ctrl + dot_slot2);
+  if (MODE == 0 && dot_slot3 >= 0)
+    block_reduce_atomic(d2, ctrl + dot_slot3);
   if (CF != CF_NONE) {
     if (fanin_last_block(ctrl) && threadIdx.x == 0)
       run_ctrl_tail(CF, ctrl, nullptr);
@@ -627,7 +635,7 @@ __global__ void k_hess_wide(const int* __restrict__ row_ptr,
                             const double* __restrict__ dotW,
                             double* __restrict__ ctrl,
                             int n, int dot_slot, int dot_slot2,
-                            int guard) {
+                            int dot_slot3, int guard) {
   if (guarded_off(ctrl, guard)) return;
   constexpr int dh = D + 1;
   constexpr int TILE = dh * R;
@@ -642,7 +650,7 @@ __global__ void k_hess_wide(const int* __restrict__ row_ptr,
   const int i = blockIdx.x * PB + slot;
   const bool live = active && i < n;
   double acc = 0.0;
-  double d0 = 0.0, d1 = 0.0;
+  double d0 = 0.0, d1 = 0.0, d2 = 0.0;
   if (live) {
     const int s0 = row_ptr[i], e0 = row_ptr[i + 1];
     for (int p = s0; p < e0; ++p) {
@@ -655,6 +663,8 @@ __global__ void k_hess_wide(const int* __restrict__ row_ptr,
     }
     const double g = G ? G[(size_t)i * TILE + e] : 0.0;
     acc += g;
+    if (MODE == 0 && dot_slot3 >= 0 && G)
+      d2 = g * X[(size_t)i * TILE + e];
     if (MODE == 1) {
       const double v = V[(size_t)i * TILE + e];
       d0 = (acc - g) * v;
@@ -692,6 +702,8 @@ __global__ void k_hess_wide(const int* __restrict__ row_ptr,
   }
   if (dot_slot >= 0) block_reduce_atomic(d0, ctrl + dot_slot);
   if (dot_slot2 >= 0) block_reduce_atomic(d1, ctrl + dot_slot2);
+  if (MODE == 0 && dot_slot3 >= 0)
+    block_reduce_atomic(d2, ctrl + dot_slot3);
   if (CF != CF_NONE) {
     if (fanin_last_block(ctrl) && threadIdx.x == 0)
       run_ctrl_tail(CF, ctrl, nullptr);
@@ -739,12 +751,13 @@ __global__ void k_bsr_spmm_mfma_d3(const int* __restrict__ grp_ptr,
         ? X[(size_t)j * 4 * r + kfrag * r + col] : 0.0;
     acc = __builtin_amdgcn_mfma_f64_16x16x4f64(a, b, acc, 0, 0, 0);
   }
-  // D layout: lane l, reg q -> D[row = 4*(l>>4)+q][col = l&15]
+  // D layout (probed on gfx950, scripts/mfma_probe.py): lane l, reg q
+  // -> D[row = 4*q + (l>>4)][col = l&15]
   const int col = lane & 15;
   if (col < r) {
     #pragma unroll
     for (int q = 0; q < 4; ++q) {
-      const int row = 4 * (lane >> 4) + q;
+      const int row = 4 * q + (lane >> 4);
       const int pose = g * 4 + (row >> 2);
       if (pose < n)
         out[(size_t)pose * 4 * r + (row & 3) * r + col] = acc[q];
@@ -2129,7 +2142,11 @@ static inline bool hess_use_wide(int n) {
   }();
   if (mode == 0) return false;
   if (mode == 1) return true;
-  return n >= 16384;
+  // Measured (profiles/r2c_*): the thread-per-pose kernel runs at
+  // ~HBM roofline even at 125k poses/agent (concurrent agents supply
+  // the missing waves), while the element-per-thread layout pays ~5x
+  // redundant Q-row reads. Wide stays opt-in for experiments.
+  return false;
 }
 
 template <int MODE, int CF = CF_NONE>
@@ -2139,7 +2156,7 @@ static void launch_hess_fused(const int* rp, const int* ci,
                               double* out, const double* dotW,
                               double* ctrl, int n, int d, int r,
                               int dot_slot, int dot_slot2, int guard,
-                              hipStream_t s) {
+                              hipStream_t s, int dot_slot3 = -1) {
   const int grid = blocks_for(n, 256);
   const bool wide = hess_use_wide(n);
 #define CASE_HF(D, R) \
@@ -2149,11 +2166,13 @@ static void launch_hess_fused(const int* rp, const int* ci,
       hipLaunchKernelGGL((k_hess_wide<D, R, MODE, CF>), \
                          dim3(blocks_for(n, PB)), dim3(256), 0, s, \
                          rp, ci, vals, V, X, G, out, dotW, \
-                         ctrl, n, dot_slot, dot_slot2, guard); \
+                         ctrl, n, dot_slot, dot_slot2, dot_slot3, \
+                         guard); \
     } else { \
       hipLaunchKernelGGL((k_hess_fused<D, R, MODE, CF>), dim3(grid), \
                          dim3(256), 0, s, rp, ci, vals, V, X, G, out, \
-                         dotW, ctrl, n, dot_slot, dot_slot2, guard); \
+                         dotW, ctrl, n, dot_slot, dot_slot2, dot_slot3, \
+                         guard); \
     } \
     return; \
   }
@@ -2730,11 +2749,8 @@ static void enqueue_solve_body(DpoCtx* c, double* X, const double* nbr,
   // two-kernel split re-read the full iterate from HBM)
   launch_hess_fused<0>(c->q_rp, c->q_ci, c->q_vals, X, X, c->Gt,
                        c->grad, nullptr, c->ctrl, n, d, r,
-                       C_DOT1, C_DOT0, -1, s);
-  if (c->Gt)
-    hipLaunchKernelGGL(k_dots, dim3(gvec), dim3(256), 0, s,
-                       c->Gt, X, (const double*)nullptr, c->ctrl,
-                       C_DOT2, -1, total, -1);
+                       C_DOT1, C_DOT0, -1, s,
+                       c->Gt ? C_DOT2 : -1);  // <G,X> folded in
   hipLaunchKernelGGL(k_axpby, dim3(gvec), dim3(256), 0, s,
                      c->grad, (const double*)nullptr, 1.0, 0.0, c->rvec,
                      total);
@@ -3041,22 +3057,14 @@ void dpo_eval_terms(void* h, const double* X, double* out_dev,
   const int gvec = (int)((total + 255) / 256);
   static const bool no_cf = dpo_env_flag("DPO_NO_CF");
   dzero(c->ctrl + C_DOT0, 4, s);
-  // fused: Q@X + G, projection at X, <P,P> and <QX+G,X> in one kernel
+  (void)gvec; (void)total; (void)no_cf;
+  // ONE fused kernel: Q@X + G, projection at X, and all three scalars
+  // <P,P> / <QX+G,X> / <G,X> (the standalone <G,X> k_dots re-read
+  // 2x the iterate from HBM and was the eval phase's fattest kernel
+  // at 1M poses — profiles/r2c_narrow)
   launch_hess_fused<0>(c->q_rp, c->q_ci, c->q_vals, X, X, c->Gt,
                        c->grad, nullptr, c->ctrl, n, d, r,
-                       C_DOT1, C_DOT0, -1, s);
-  if (c->Gt && !no_cf) {
-    // the <G, X> dot's last block combines all three scalars into
-    // out_dev (CF_COMBINE tail) instead of a separate kernel
-    hipLaunchKernelGGL((k_dots<CF_COMBINE>), dim3(gvec), dim3(256), 0, s,
-                       c->Gt, X, (const double*)nullptr, c->ctrl,
-                       C_DOT2, -1, total, -1, out_dev);
-    return;
-  }
-  if (c->Gt)
-    hipLaunchKernelGGL(k_dots, dim3(gvec), dim3(256), 0, s,
-                       c->Gt, X, (const double*)nullptr, c->ctrl,
-                       C_DOT2, -1, total, -1);
+                       C_DOT1, C_DOT0, -1, s, c->Gt ? C_DOT2 : -1);
   hipLaunchKernelGGL(k_eval_combine, dim3(1), dim3(64), 0, s, c->ctrl,
                      out_dev);
 }
