@@ -501,3 +501,20 @@ def test_chordal_soa_gpu_matches_cpu():
     assert np.abs(T_cpu - T_gpu).max() < 1e-6
     T_ref = chordal_initialization(3, n, ma.to_list())
     assert np.abs(T_gpu - T_ref).max() < 1e-6
+
+
+def test_bsr_spmm_mfma_matches(grid_fixture):
+    """fp64-MFMA grouped-ELL SpMM (A/B study kernel) must match the
+    production BSR SpMM exactly (same fp64 FMA semantics)."""
+    from dpo_amd.ops import hip_backend as hb
+    meas, n, d, r, Q, X, V = grid_fixture
+    Qd = Q.to(DEV)
+    Xd = X.to(DEV)
+    ref = Qd.spmm(Xd)
+    gp, gc, gb = hb.build_spmm_mfma_groups(Qd.row_ptr, Qd.col_idx, n)
+    out = hb.bsr_spmm_mfma(torch.from_numpy(gp).to(DEV),
+                           torch.from_numpy(gc).to(DEV),
+                           torch.from_numpy(gb).to(DEV),
+                           Qd.vals, Xd)
+    torch.cuda.synchronize()
+    assert torch.allclose(out.cpu(), ref.cpu(), atol=1e-10)
