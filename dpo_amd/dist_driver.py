@@ -803,11 +803,21 @@ class DistributedRBCDDriver:
                     group.ids([self._group_lidx[rb] for rb in ca
                                if rb in self.local_agents])
                     for ca in color_active]
-        chunk = 16 if (gradnorm_tol <= 0.0
-                       and self.selection not in ("greedy",
-                                                  "colored_greedy")
-                       and _os_.environ.get("DPO_DRIVER_EVAL_GRAPH",
-                                            "0") != "1") else 1
+        # Pipelined eval readback width. tol<=0: nothing to decide per
+        # round -> deep ring. tol>0 with a selection that needs no
+        # per-round argmax (colored): still pipeline, checking the
+        # convergence test on ring flushes — the crossing ITERATION is
+        # read exactly from the per-round ring entries, only the break
+        # happens up to chunk-1 rounds later (extra monotone rounds).
+        if _os_.environ.get("DPO_DRIVER_EVAL_GRAPH", "0") == "1":
+            chunk = 1
+        elif gradnorm_tol <= 0.0 and self.selection not in (
+                "greedy", "colored_greedy"):
+            chunk = 16
+        elif self.selection == "colored":
+            chunk = 8
+        else:
+            chunk = 1
         evalring = (torch.zeros(chunk, self.num_robots, 3,
                                 dtype=torch.float64, device=dev)
                     if chunk > 1 else None)
@@ -933,9 +943,19 @@ class DistributedRBCDDriver:
                     ph[3] += time.perf_counter() - tD
                 if pending == chunk or it + 1 == max_iters:
                     tE = time.perf_counter()
+                    scan_from = len(res.trace)
                     flush_ring()
                     if _timing:
                         ph[4] += time.perf_counter() - tE
+                    if gradnorm_tol > 0.0:
+                        for k in range(scan_from, len(res.trace)):
+                            if res.trace[k][1] < gradnorm_tol:
+                                res.converged = True
+                                res.iterations = k + 1
+                                res.trace = res.trace[:k + 1]
+                                break
+                        if res.converged:
+                            break
                     if time_limit_s and \
                             time.perf_counter() - t0 > time_limit_s:
                         break
