@@ -711,8 +711,15 @@ class DistributedRBCDDriver:
         dev = pk["dev"]
         selected = 0
         t0 = time.perf_counter()
-        evalmat = torch.zeros(self.num_robots, 3, dtype=torch.float64,
-                              device=dev)
+        # persistent eval buffers: stable pointers keep the native
+        # gather/eval graph caches hot across repeated run() calls
+        # (bench episodes)
+        if getattr(self, "_evalmat", None) is None \
+                or self._evalmat.device != dev:
+            self._evalmat = torch.zeros(self.num_robots, 3,
+                                        dtype=torch.float64, device=dev)
+        evalmat = self._evalmat
+        evalmat.zero_()
         sizes = pk["rank_payload_len"]
         accel = self.acceleration
         # Nesterov host-side scalars per agent
@@ -818,9 +825,16 @@ class DistributedRBCDDriver:
             chunk = 8
         else:
             chunk = 1
-        evalring = (torch.zeros(chunk, self.num_robots, 3,
-                                dtype=torch.float64, device=dev)
-                    if chunk > 1 else None)
+        if chunk > 1:
+            er = getattr(self, "_evalring", None)
+            if er is None or er.shape[0] != chunk:
+                self._evalring = torch.zeros(chunk, self.num_robots, 3,
+                                             dtype=torch.float64,
+                                             device=dev)
+            evalring = self._evalring
+            evalring.zero_()
+        else:
+            evalring = None
         pending = 0   # rounds evaluated but not yet read back
 
         def flush_ring():
