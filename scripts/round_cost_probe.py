@@ -22,6 +22,8 @@ def main():
     ap.add_argument("--selection", default="colored")
     ap.add_argument("--segments", type=int, default=10)
     ap.add_argument("--seglen", type=int, default=25)
+    ap.add_argument("--inner-tol", type=float, default=1e-2)
+    ap.add_argument("--partition", default="contiguous")
     args = ap.parse_args()
     from dpo_amd.comm import init_from_env
     from dpo_amd.dist_driver import DistributedRBCDDriver
@@ -29,8 +31,10 @@ def main():
     meas, n = load_dataset(args.dataset)
     comm = init_from_env(args.device)
     drv = DistributedRBCDDriver(meas, n, args.agents, comm, r=5,
-                                partition="contiguous", device=args.device,
-                                selection=args.selection)
+                                partition=args.partition,
+                                device=args.device,
+                                selection=args.selection,
+                                inner_tol=args.inner_tol)
     drv.snapshot_initial_state()
     # warmup episode (graph capture etc.)
     drv.run(max_iters=40, gradnorm_tol=0.0)
