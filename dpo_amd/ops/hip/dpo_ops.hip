@@ -2734,7 +2734,15 @@ static void enqueue_solve_body(DpoCtx* c, double* X, const double* nbr,
   // per-stage launch path's j-split apply sustains far higher
   // bandwidth (measured: city10000/cubicle agents at N ~ 4600-6000
   // were 4-8x slower under the persistent path).
-  const bool persist_ok = !no_persist && !no_cf && c->Minv != nullptr
+  // Round 2: the persistent kernel is now OPT-IN (DPO_PERSIST=1).
+  // After the gradient/eval fusion cut the staged path's launch count,
+  // A/B probes (gpurun_out/r2h_probe.log) measured the staged path at
+  // ~1.4 ms/round vs ~2.6 ms/round persistent on the 8-agent sphere2500
+  // bench: at 25-workgroup grids the in-kernel grid barriers cost
+  // 380-850 us per solve loop under multi-agent concurrency.
+  static const bool persist_opt_in = dpo_env_flag("DPO_PERSIST");
+  const bool persist_ok = persist_opt_in && !no_persist && !no_cf
+      && c->Minv != nullptr
       && c->N <= 1500 && (total + 255) / 256 <= 256;
   if (persist_ok && persist_full) {
     launch_solve_persist(c, X, nbr, tol, Delta0, accept_rho, s, 1);
