@@ -415,6 +415,96 @@ __global__ void k_proj_dots(const double* __restrict__ X,
   }
 }
 
+// Fused block-Jacobi preconditioner apply + tangent projection (+ dot
+// + control tail): one kernel instead of the {precond, proj} pair —
+// one of five serial stages in every tCG iteration, so at large N the
+// removed launch+drain latency is pure win. Per pose: y = M^-1 r via
+// the cached per-pose Cholesky factors, P = P_X(y), dot <P, r>.
+template <int D, int R, int CF = CF_NONE>
+__global__ void k_precond_proj(const double* __restrict__ L,
+                               const double* __restrict__ rvec,
+                               const double* __restrict__ X,
+                               double* __restrict__ Z,
+                               double* __restrict__ ctrl,
+                               int n, int dot_slot, int guard) {
+  if (guarded_off(ctrl, guard)) return;
+  constexpr int dh = D + 1;
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  double Vt[dh][R];
+  double dot = 0.0;
+  if (i < n) {
+    const double* Li = L + (size_t)i * dh * dh;
+    const double* Ri = rvec + (size_t)i * dh * R;
+    // per-column forward/backward triangular solves (L L^T y = r)
+    #pragma unroll
+    for (int k = 0; k < R; ++k) {
+      double y[dh];
+      #pragma unroll
+      for (int a = 0; a < dh; ++a) {
+        double sv = Ri[a * R + k];
+        #pragma unroll
+        for (int b = 0; b < dh; ++b)
+          if (b < a) sv = fma(-Li[a * dh + b], y[b], sv);
+        y[a] = sv / Li[a * dh + a];
+      }
+      #pragma unroll
+      for (int a = dh - 1; a >= 0; --a) {
+        double sv = y[a];
+        #pragma unroll
+        for (int b = 0; b < dh; ++b)
+          if (b > a) sv = fma(-Li[b * dh + a], y[b], sv);
+        y[a] = sv / Li[a * dh + a];
+      }
+      #pragma unroll
+      for (int a = 0; a < dh; ++a) Vt[a][k] = y[a];
+    }
+    // tangent projection at X (Stiefel rows only)
+    const double* Xi = X + (size_t)i * dh * R;
+    double S[D][D];
+    #pragma unroll
+    for (int a = 0; a < D; ++a)
+      #pragma unroll
+      for (int b = 0; b < D; ++b) {
+        double sv = 0.0;
+        #pragma unroll
+        for (int k = 0; k < R; ++k)
+          sv = fma(Xi[a * R + k], Vt[b][k], sv);
+        S[a][b] = sv;
+      }
+    #pragma unroll
+    for (int a = 0; a < D; ++a)
+      #pragma unroll
+      for (int b = a; b < D; ++b) {
+        const double sv = 0.5 * (S[a][b] + S[b][a]);
+        S[a][b] = sv;
+        S[b][a] = sv;
+      }
+    #pragma unroll
+    for (int a = 0; a < D; ++a)
+      #pragma unroll
+      for (int k = 0; k < R; ++k) {
+        double v = Vt[a][k];
+        #pragma unroll
+        for (int b = 0; b < D; ++b) v = fma(-S[a][b], Xi[b * R + k], v);
+        Vt[a][k] = v;
+      }
+    double* Zi = Z + (size_t)i * dh * R;
+    #pragma unroll
+    for (int c = 0; c < dh; ++c)
+      #pragma unroll
+      for (int k = 0; k < R; ++k) {
+        Zi[c * R + k] = Vt[c][k];
+        if (dot_slot >= 0)
+          dot = fma(Vt[c][k], Ri[c * R + k], dot);
+      }
+  }
+  if (dot_slot >= 0) block_reduce_atomic(dot, ctrl + dot_slot);
+  if (CF != CF_NONE) {
+    if (fanin_last_block(ctrl) && threadIdx.x == 0)
+      run_ctrl_tail(CF, ctrl, nullptr);
+  }
+}
+
 // Wide (element-per-thread) tangent projection for LARGE agents: same
 // wave-starvation fix as k_hess_wide (a thread-per-pose kernel gets
 // only n/64 waves). dh*r threads per pose; the cross-element sym(Y V^T)
@@ -2574,6 +2664,29 @@ static void ctx_precond(DpoCtx* c, const double* V, double* Z,
                           ST_RUN, s);
 }
 
+// Fused z-stage: z = P_X(M^-1 r) + <z, r> dot + CF tail in one kernel
+// when the block-Jacobi preconditioner is active (large agents);
+// otherwise the dense apply runs standalone followed by the fused
+// projection kernel, as before.
+template <int CF>
+static void ctx_precond_proj(DpoCtx* c, const double* X, hipStream_t s) {
+  if (c->Minv == nullptr) {
+    const int grid = blocks_for(c->n, 256);
+#define CASE_PP(D, R) \
+    if (c->d == D && c->r == R) { \
+      hipLaunchKernelGGL((k_precond_proj<D, R, CF>), dim3(grid), \
+                         dim3(256), 0, s, c->Ljac, c->rvec, X, c->z, \
+                         c->ctrl, c->n, C_DOT0, ST_RUN); \
+      return; \
+    }
+    DPO_FOREACH_DR(CASE_PP)
+#undef CASE_PP
+  }
+  ctx_precond(c, c->rvec, c->z, s);
+  launch_proj_dots_cf<CF>(X, c->z, nullptr, c->z, c->rvec, c->ctrl,
+                          c->n, c->d, c->r, C_DOT0, -1, ST_RUN, s);
+}
+
 static void ctx_spmm(DpoCtx* c, const double* X, double* out, int guard,
                      hipStream_t s) {
   launch_spmm(c->q_rp, c->q_ci, c->q_vals, c->n, c->d, c->r, X, out,
@@ -2765,14 +2878,13 @@ static void enqueue_solve_body(DpoCtx* c, double* X, const double* nbr,
   hipLaunchKernelGGL(k_ctrl_init, dim3(1), dim3(64), 0, s, c->ctrl, tol,
                      Delta0, 1.0, 0.1);
   // z0
-  ctx_precond(c, c->rvec, c->z, s);
   if (no_cf) {
+    ctx_precond(c, c->rvec, c->z, s);
     launch_proj_dots(X, c->z, nullptr, c->z, c->rvec, c->ctrl, n, d, r,
                      C_DOT0, -1, ST_RUN, s);
     hipLaunchKernelGGL(k_ctrl_z0, dim3(1), dim3(64), 0, s, c->ctrl);
   } else {
-    launch_proj_dots_cf<CF_Z0>(X, c->z, nullptr, c->z, c->rvec, c->ctrl,
-                               n, d, r, C_DOT0, -1, ST_RUN, s);
+    ctx_precond_proj<CF_Z0>(c, X, s);
   }
   hipLaunchKernelGGL(k_tcg_delta, dim3(gvec), dim3(256), 0, s,
                      c->delta, c->z, c->ctrl, total);
@@ -2803,15 +2915,13 @@ static void enqueue_solve_body(DpoCtx* c, double* X, const double* nbr,
                          c->Hd, c->eta_snap, c->delta_snap, c->ctrl,
                          total);
     }
-    ctx_precond(c, c->rvec, c->z, s);
     if (no_cf) {
+      ctx_precond(c, c->rvec, c->z, s);
       launch_proj_dots(X, c->z, nullptr, c->z, c->rvec, c->ctrl, n, d, r,
                        C_DOT0, -1, ST_RUN, s);
       hipLaunchKernelGGL(k_ctrl_beta, dim3(1), dim3(64), 0, s, c->ctrl);
     } else {
-      launch_proj_dots_cf<CF_BETA>(X, c->z, nullptr, c->z, c->rvec,
-                                   c->ctrl, n, d, r, C_DOT0, -1, ST_RUN,
-                                   s);
+      ctx_precond_proj<CF_BETA>(c, X, s);
     }
     hipLaunchKernelGGL(k_tcg_delta, dim3(gvec), dim3(256), 0, s,
                        c->delta, c->z, c->ctrl, total);
