@@ -580,9 +580,10 @@ class PGOAgent:
                 self._dev_solver = DeviceSolver(self.n, self.d, self.r,
                                                 self.device, max_inner=10)
             X_work = X_start.clone() if acceleration else self.X
-            stats = self._dev_solver.solve(self.problem, X_work,
-                                           tol=self.params.inner_tol,
-                                           Delta0=100.0)
+            for _ in range(self.params.tr_max_iterations):
+                stats = self._dev_solver.solve(self.problem, X_work,
+                                               tol=self.params.inner_tol,
+                                               Delta0=100.0)
             self.X = X_work
             from .types import OptResult
             res = OptResult(success=True,
@@ -592,7 +593,8 @@ class PGOAgent:
             self.last_opt_result = res
             return True
         tr = TRParams(tolerance=self.params.inner_tol, initial_radius=100.0,
-                      max_iterations=1, max_inner_iterations=10)
+                      max_iterations=self.params.tr_max_iterations,
+                      max_inner_iterations=10)
         opt = QuadraticOptimizer(self.problem, self.params.algorithm, tr,
                                  verbose=self.params.verbose)
         self.X = opt.optimize(X_start)
@@ -976,8 +978,10 @@ class PGOAgent:
         if accel:
             self.X.copy_(self.Y)
         nbr = self._nbr_buffer_aux if accel else self._nbr_buffer
-        self._dev_solver.round_solve(self.X, nbr, tol=self.params.inner_tol,
-                                     Delta0=100.0)
+        for _ in range(self.params.tr_max_iterations):
+            self._dev_solver.round_solve(self.X, nbr,
+                                         tol=self.params.inner_tol,
+                                         Delta0=100.0)
 
     def _packed_eval(self, out=None):
         """Device 3-vector [f, 0.5<X,G>, ||rgrad||^2] with fresh G.
@@ -988,8 +992,8 @@ class PGOAgent:
             return out
         return self._dev_solver.round_eval(self.X, self._nbr_buffer)
 
-    def _packed_solve_async(self, accel: bool) -> None:
-        if accel:
+    def _packed_solve_async(self, accel: bool, first: bool = True) -> None:
+        if accel and first:
             self.X.copy_(self.Y)
         nbr = self._nbr_buffer_aux if accel else self._nbr_buffer
         self._dev_solver.round_solve_async(self.X, nbr,

@@ -62,7 +62,8 @@ class DistributedRBCDDriver:
                  device: str = "cpu",
                  verbose: bool = False,
                  selection: str = "greedy",
-                 inner_tol: float = 1e-2):
+                 inner_tol: float = 1e-2,
+                 tr_max_iterations: int = 1):
         self.comm = comm
         self.num_robots = num_robots
         self.verbose = verbose
@@ -198,7 +199,8 @@ class DistributedRBCDDriver:
                                acceleration=acceleration,
                                robust_cost_type=robust,
                                verbose=verbose, device=device,
-                               inner_tol=inner_tol)
+                               inner_tol=inner_tol,
+                               tr_max_iterations=tr_max_iterations)
             a = PGOAgent(rb, p)
             a.set_lifting_matrix(YL)
             T_init = None
@@ -800,6 +802,8 @@ class DistributedRBCDDriver:
                            and not _sync_solve)
         inner_tol = next(iter(self.local_agents.values())) \
             .params.inner_tol if self.local_agents else 1e-2
+        tr_steps = next(iter(self.local_agents.values())) \
+            .params.tr_max_iterations if self.local_agents else 1
         if self.selection in ("colored", "colored_greedy"):
             color_active = [
                 [rb for rb in range(self.num_robots)
@@ -909,27 +913,31 @@ class DistributedRBCDDriver:
                     gids = group.ids([self._group_lidx[selected]])
                 else:
                     gids = group.ids([])
-                if len(gids):
-                    group.solve_start(gids, tol=inner_tol)
                 if _timing:
                     tB = time.perf_counter()
                 if len(gids):
-                    group.solve_finish(gids)
+                    for _k in range(tr_steps):
+                        group.solve_start(gids, tol=inner_tol)
+                        group.solve_finish(gids)
             else:
                 for rb, a in self.local_agents.items():
-                    if rb in active:
-                        if _sync_solve:
-                            a._packed_solve(accel)
-                        else:
-                            a._packed_solve_async(accel)
-                    elif accel:
+                    if accel and rb not in active:
                         a.X.copy_(a.Y)
                 if _timing:
                     tB = time.perf_counter()
-                if not _sync_solve:
-                    for rb in active:
-                        if rb in self.local_agents:
-                            self.local_agents[rb]._packed_solve_finish()
+                if _sync_solve:
+                    for rb, a in self.local_agents.items():
+                        if rb in active:
+                            a._packed_solve(accel)  # loops tr steps
+                else:
+                    for _k in range(tr_steps):
+                        for rb in active:
+                            if rb in self.local_agents:
+                                self.local_agents[rb]._packed_solve_async(
+                                    accel, first=(_k == 0))
+                        for rb in active:
+                            if rb in self.local_agents:
+                                self.local_agents[rb]._packed_solve_finish()
             if _timing:
                 tC = time.perf_counter()
             if accel:

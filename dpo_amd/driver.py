@@ -60,7 +60,8 @@ class MultiRobotDriver:
                  robust_inner_iters: int = 30,
                  device: str = "cpu",
                  verbose: bool = False,
-                 selection: str = "greedy"):
+                 selection: str = "greedy",
+                 tr_max_iterations: int = 1):
         self.num_robots = num_robots
         self.verbose = verbose
         self.selection = selection
@@ -117,7 +118,8 @@ class MultiRobotDriver:
                                acceleration=acceleration,
                                robust_cost_type=robust,
                                robust_opt_inner_iters=robust_inner_iters,
-                               verbose=verbose, device=device)
+                               verbose=verbose, device=device,
+                               tr_max_iterations=tr_max_iterations)
             if robust_params is not None:
                 p.robust_cost_params = robust_params
             a = PGOAgent(rb, p)

@@ -140,6 +140,13 @@ class PGOAgentParams:
     # hardwires 1e-2 in PGOAgent::optimize's RBCD knobs). 0.0 forces a
     # full solve every round (benchmark mode: no converged no-op rounds).
     inner_tol: float = 1e-2
+    # dpo_amd extension: trust-region steps per RBCD round. The
+    # reference hardwires Max_Iteration=1 (one TR step per iterate,
+    # QuadraticOptimizer.cpp:92-110); >1 runs that many TR steps
+    # against the same fixed neighbor data each round — measured to
+    # cut iterations-to-convergence on weakly-coupled graphs
+    # (RESULTS.md round 2) at slightly higher per-round cost.
+    tr_max_iterations: int = 1
     verbose: bool = False
     log_data: bool = False
     log_directory: str = ""

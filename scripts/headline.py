@@ -25,6 +25,8 @@ def main():
     ap.add_argument("--partition", default="contiguous")
     ap.add_argument("--selection", default="greedy")
     ap.add_argument("--accel", action="store_true")
+    ap.add_argument("--tr-iters", type=int, default=1,
+                    help="trust-region steps per RBCD round (reference: 1)")
     ap.add_argument("--max-iters", type=int, default=1000)
     ap.add_argument("--tol", type=float, default=0.1)
     ap.add_argument("--trace", default=None)
@@ -45,7 +47,8 @@ def main():
         drv = DistributedRBCDDriver(
             meas, n, args.robots, comm, r=args.r,
             partition=args.partition, acceleration=args.accel,
-            device=args.device, selection=args.selection)
+            device=args.device, selection=args.selection,
+            tr_max_iterations=args.tr_iters)
     else:
         from dpo_amd.driver import MultiRobotDriver
         drv = MultiRobotDriver(

@@ -18,6 +18,8 @@ def main():
     ap.add_argument("--robots", type=int, default=5)
     ap.add_argument("--selection", default="greedy")
     ap.add_argument("--accel", action="store_true")
+    ap.add_argument("--tr-iters", type=int, default=1,
+                    help="trust-region steps per RBCD round (reference: 1)")
     ap.add_argument("--max-iters", type=int, default=1000)
     ap.add_argument("--datasets", default="",
                     help="comma-separated subset (default: all)")
@@ -39,7 +41,8 @@ def main():
             drv = DistributedRBCDDriver(
                 meas, n, args.robots, Comm(), r=5,
                 partition=args.partition, device=args.device,
-                selection=args.selection, acceleration=args.accel)
+                selection=args.selection, acceleration=args.accel,
+                tr_max_iterations=args.tr_iters)
             res = drv.run(max_iters=args.max_iters)
             print(json.dumps({
                 "dataset": name, "poses": n, "edges": len(meas),
