@@ -69,7 +69,7 @@ def main():
         from dpo_amd.chordal import chordal_initialization_soa
         t0c = time.perf_counter()
         ma.warm_start = chordal_initialization_soa(
-            ma, n, device=args.device, tol=1e-6, max_iters=400)
+            ma, n, device=args.device, tol=1e-6, max_iters=4000)
         print(f"# chordal init (SoA/GPU): "
               f"{time.perf_counter() - t0c:.2f}s", file=sys.stderr)
     comm = init_from_env(args.device)
