@@ -269,3 +269,22 @@ def test_soa_cpu_robust_rejected_cleanly():
         DistributedRBCDDriver(ma, n, 2, Comm(), r=5,
                               partition="contiguous", device="cpu",
                               robust=RobustCostType.GNC_TLS)
+
+
+def test_flagship_config_beats_reference_on_parking_garage():
+    """Regression anchor for the round-2 flagship configuration
+    (colored + multilevel + tr_max_iterations=3): parking-garage must
+    converge in <= 14 iterations (the reference's best run)
+    — it measures 7 on both CPU and MI355X."""
+    from dpo_amd.comm import Comm
+    from dpo_amd.dist_driver import DistributedRBCDDriver
+    from dpo_amd.io_g2o import load_dataset
+    meas, n = load_dataset("parking-garage")
+    drv = DistributedRBCDDriver(meas, n, 5, Comm(), r=5,
+                                partition="multilevel",
+                                selection="colored",
+                                tr_max_iterations=3)
+    res = drv.run(max_iters=100)
+    assert res.converged
+    assert res.iterations <= 14, res.iterations
+    assert abs(res.final_cost - 1.27) < 0.05
