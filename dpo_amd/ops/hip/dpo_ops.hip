@@ -1095,13 +1095,13 @@ __global__ void k_tcg_update(double* __restrict__ eta,
                              double* __restrict__ ctrl,
                              long total) {
   if (guarded_off(ctrl, ST_RUN)) return;
-  const long i0 = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 2;
+  const long i0 = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
   const double coef = ctrl[C_COEF];
   const int stop_pending = (int)ctrl[C_STOP_PENDING];
   const int j = (int)ctrl[C_ITER];
   double rr = 0.0;
   #pragma unroll
-  for (int u = 0; u < 2; ++u) {
+  for (int u = 0; u < 4; ++u) {
     const long i = i0 + u;
     if (i < total) {
       const double dl = delta[i];
@@ -2382,7 +2382,7 @@ void dpo_tcg_update(double* eta, double* rvec, const double* delta,
                     double* ctrl, long total, void* stream) {
   hipStream_t s = (hipStream_t)stream;
   hipLaunchKernelGGL((k_tcg_update<CF_NONE>),
-                     dim3(blocks_for((total + 1) / 2, 256)),
+                     dim3(blocks_for((total + 3) / 4, 256)),
                      dim3(256), 0, s, eta, rvec, delta, Hd, eta_snap,
                      delta_snap, ctrl, total);
 }
@@ -2900,7 +2900,7 @@ static void enqueue_solve_body(DpoCtx* c, double* X, const double* nbr,
                            C_DOT0, -1, ST_RUN, s);
       hipLaunchKernelGGL(k_ctrl_alpha, dim3(1), dim3(64), 0, s, c->ctrl);
       hipLaunchKernelGGL((k_tcg_update<CF_NONE>),
-                         dim3(blocks_for((total + 1) / 2, 256)),
+                         dim3(blocks_for((total + 3) / 4, 256)),
                          dim3(256), 0, s, c->eta, c->rvec, c->delta,
                          c->Hd, c->eta_snap, c->delta_snap, c->ctrl,
                          total);
@@ -2910,7 +2910,7 @@ static void enqueue_solve_body(DpoCtx* c, double* X, const double* nbr,
           c->q_rp, c->q_ci, c->q_vals, c->delta, X, nullptr, c->Hd,
           c->delta, c->ctrl, n, d, r, C_DOT0, -1, ST_RUN, s);
       hipLaunchKernelGGL((k_tcg_update<CF_RR>),
-                         dim3(blocks_for((total + 1) / 2, 256)),
+                         dim3(blocks_for((total + 3) / 4, 256)),
                          dim3(256), 0, s, c->eta, c->rvec, c->delta,
                          c->Hd, c->eta_snap, c->delta_snap, c->ctrl,
                          total);
