@@ -89,3 +89,24 @@ def test_world4_accel_colored_combined_payload():
     assert out["iters"] == ref.iterations
     assert abs(out["cost"] - ref.final_cost) \
         < 1e-6 * max(1, abs(ref.final_cost))
+
+
+def test_bench_torchrun_world2_cpu():
+    """bench.py under torch.distributed.run with 2 ranks (gloo, CPU):
+    the exact launch path the driver's multi-GPU SCALE run uses. The
+    episode must converge with the same iteration count as world=1."""
+    import subprocess
+    import sys
+    here = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29787", "bench.py", "--gpus", "2",
+         "--steps", "2", "--warmup", "1", "--dataset", "smallGrid3D",
+         "--agents", "4"],
+        capture_output=True, text=True, timeout=600, cwd=here)
+    assert out.returncode == 0, out.stderr[-2000:]
+    o = json.loads(out.stdout.strip().splitlines()[-1])
+    assert o["n_gpus"] == 2
+    assert o["config"]["converged"] is True
+    assert o["config"]["iterations"] == 56  # world=1 count (smallGrid3D)
