@@ -288,3 +288,14 @@ def test_flagship_config_beats_reference_on_parking_garage():
     assert res.converged
     assert res.iterations <= 14, res.iterations
     assert abs(res.final_cost - 1.27) < 0.05
+
+
+def test_colored_greedy_selection_converges():
+    from dpo_amd.comm import Comm
+    from dpo_amd.dist_driver import DistributedRBCDDriver
+    meas, n = grid3d(side=4, seed=0)
+    drv = DistributedRBCDDriver(meas, n, 3, Comm(), r=5,
+                                partition="contiguous",
+                                selection="colored_greedy")
+    res = drv.run(max_iters=400)
+    assert res.converged

@@ -518,3 +518,20 @@ def test_bsr_spmm_mfma_matches(grid_fixture):
                            Qd.vals, Xd)
     torch.cuda.synchronize()
     assert torch.allclose(out.cpu(), ref.cpu(), atol=1e-10)
+
+
+def test_flagship_tr3_gpu_beats_reference_parking_garage():
+    """GPU mirror of the CPU flagship regression: colored + multilevel
+    + tr_max_iterations=3 must beat the reference's best parking-garage
+    run (14 iterations) through the packed device path."""
+    from dpo_amd.comm import Comm
+    from dpo_amd.dist_driver import DistributedRBCDDriver
+    from dpo_amd.io_g2o import load_dataset
+    meas, n = load_dataset("parking-garage")
+    drv = DistributedRBCDDriver(meas, n, 5, Comm(), r=5,
+                                partition="multilevel",
+                                selection="colored", device=DEV,
+                                tr_max_iterations=3)
+    res = drv.run(max_iters=100)
+    assert res.converged
+    assert res.iterations <= 14, res.iterations
