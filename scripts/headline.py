@@ -54,7 +54,8 @@ def main():
         drv = MultiRobotDriver(
             meas, n, args.robots, r=args.r, partition=args.partition,
             acceleration=args.accel, device=args.device,
-            selection=args.selection)
+            selection=args.selection,
+            tr_max_iterations=args.tr_iters)
     setup_s = time.perf_counter() - t_setup
     res = drv.run(max_iters=args.max_iters, gradnorm_tol=args.tol,
                   trace_file=args.trace)
