@@ -2511,13 +2511,23 @@ void dpo_gnc_weights(const double* X, const double* nbr, const long* e1_idx,
 // =====================================================================
 
 // combine kernel: out[0..2] = [f, 0.5<X,G>, gn2] from dot slots
-__global__ void k_eval_combine(const double* __restrict__ ctrl,
+__global__ void k_eval_combine(double* __restrict__ ctrl,
                                double* __restrict__ out) {
   if (threadIdx.x != 0) return;
   // C_DOT0 = <QX + G, X>, C_DOT2 = <G, X>, C_DOT1 = ||P_X(QX+G)||^2
   out[0] = 0.5 * (ctrl[C_DOT0] + ctrl[C_DOT2]);   // f(X)
   out[1] = 0.5 * ctrl[C_DOT2];                    // 0.5 <X, G>
   out[2] = ctrl[C_DOT1];                          // gradnorm^2
+  // self-clean the dot slots for the next eval: saves the standalone
+  // 32-byte dzero launch per evaluation (12.8% of bench-scale kernel
+  // launches were k_dzero calls, profiles/bench_r2_kernel_stats.csv).
+  // Solves re-zero the whole control block themselves; the ctx
+  // allocation is zero-initialized, so the invariant "dot slots are
+  // zero on eval entry" holds from the start.
+  ctrl[C_DOT0] = 0.0;
+  ctrl[C_DOT1] = 0.0;
+  ctrl[C_DOT2] = 0.0;
+  ctrl[C_DOT3] = 0.0;
 }
 
 struct DpoCtx {
@@ -3174,8 +3184,9 @@ void dpo_eval_terms(void* h, const double* X, double* out_dev,
   const long total = c->total;
   const int gvec = (int)((total + 255) / 256);
   static const bool no_cf = dpo_env_flag("DPO_NO_CF");
-  dzero(c->ctrl + C_DOT0, 4, s);
   (void)gvec; (void)total; (void)no_cf;
+  // dot slots are pre-zeroed (zero-init at allocation; k_eval_combine
+  // self-cleans after each eval; solves wipe the whole control block)
   // ONE fused kernel: Q@X + G, projection at X, and all three scalars
   // <P,P> / <QX+G,X> / <G,X> (the standalone <G,X> k_dots re-read
   // 2x the iterate from HBM and was the eval phase's fattest kernel
